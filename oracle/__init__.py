@@ -1,0 +1,180 @@
+"""oracle — ctypes wrapper over the CPU oracle (TEST INFRASTRUCTURE ONLY).
+
+Scope (see ec_ref.h): this package is the parity checker for the MI355X EC
+backend. It may be imported only from tests/, __graft_entry__.smoke() (as the
+checker) and bench.py's cpu_baseline leg. It is NOT the product path.
+"""
+import ctypes
+import os
+import subprocess
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+
+# Technique ids (must match ec_ref.h / include/ec_mi355x.h)
+T_RS_VAN_ISA = 0
+T_CAUCHY_ISA = 1
+T_RS_VAN_JERASURE = 2
+TECHNIQUES = {
+    "reed_sol_van": T_RS_VAN_ISA,
+    "cauchy": T_CAUCHY_ISA,
+    "jerasure_reed_sol_van": T_RS_VAN_JERASURE,
+}
+
+
+def _build_if_needed():
+    for so, srcs in (("libec_ref.so", ["ec_ref.c"]),
+                     ("libec_cpu.so", ["ec_cpu.c", "ec_ref.c"])):
+        sop = os.path.join(_DIR, so)
+        if not os.path.exists(sop) or any(
+                os.path.getmtime(os.path.join(_DIR, s)) > os.path.getmtime(sop)
+                for s in srcs):
+            subprocess.run(["make", "-C", _DIR], check=True,
+                           capture_output=True)
+            break
+
+
+def _load(name):
+    _build_if_needed()
+    return ctypes.CDLL(os.path.join(_DIR, name))
+
+
+_ref = _load("libec_ref.so")
+_cpu = _load("libec_cpu.so")
+
+_ref.ecref_gf_init()
+_ref.ecref_gf_mul.restype = ctypes.c_uint8
+_ref.ecref_gf_mul.argtypes = [ctypes.c_uint8, ctypes.c_uint8]
+_ref.ecref_gf_inv.restype = ctypes.c_uint8
+_ref.ecref_gf_inv.argtypes = [ctypes.c_uint8]
+_ref.ecref_gf_log_table.restype = ctypes.POINTER(ctypes.c_uint8 * 256)
+_ref.ecref_gf_exp_table.restype = ctypes.POINTER(ctypes.c_uint8 * 256)
+_ref.ecref_matrix.restype = ctypes.c_int
+_ref.ecref_matrix.argtypes = [ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
+                              ctypes.c_int]
+_ref.ecref_decode.restype = ctypes.c_int
+_ref.ecref_chunk_size_isa.restype = ctypes.c_uint
+_ref.ecref_chunk_size_isa.argtypes = [ctypes.c_int, ctypes.c_uint]
+_ref.ecref_chunk_size_jerasure.restype = ctypes.c_uint
+_ref.ecref_chunk_size_jerasure.argtypes = [ctypes.c_int, ctypes.c_int,
+                                           ctypes.c_uint]
+_cpu.eccpu_encode_batch.restype = ctypes.c_int
+_cpu.eccpu_decode_batch.restype = ctypes.c_int
+_cpu.eccpu_threads.restype = ctypes.c_int
+
+
+def gf_mul(a, b):
+    return _ref.ecref_gf_mul(a, b)
+
+
+def gf_inv(a):
+    return _ref.ecref_gf_inv(a)
+
+
+def gf_log_table():
+    return np.frombuffer(bytes(_ref.ecref_gf_log_table().contents), np.uint8)
+
+
+def gf_exp_table():
+    return np.frombuffer(bytes(_ref.ecref_gf_exp_table().contents), np.uint8)
+
+
+def matrix(technique, k, m):
+    """Full (k+m) x k generator (identity top), as uint8 ndarray."""
+    t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    a = np.zeros((k + m, k), dtype=np.uint8)
+    r = _ref.ecref_matrix(t, a.ctypes.data_as(ctypes.c_void_p), k, m)
+    if r != 0:
+        raise ValueError(f"ecref_matrix failed: {r}")
+    return a
+
+
+def _ptr_array(bufs):
+    arr = (ctypes.c_void_p * len(bufs))()
+    for i, b in enumerate(bufs):
+        arr[i] = None if b is None else b.ctypes.data_as(ctypes.c_void_p).value
+    return arr
+
+
+def encode(technique, k, m, data, chunk_bytes=None):
+    """data: list of k uint8 arrays (or None for zeros). Returns list of m
+    parity arrays. Scalar oracle (ecref_encode)."""
+    lens = {d.nbytes for d in data if d is not None}
+    assert len(lens) == 1
+    length = lens.pop()
+    gen = matrix(technique, k, m)
+    rows = np.ascontiguousarray(gen[k:])
+    parity = [np.zeros(length, dtype=np.uint8) for _ in range(m)]
+    _ref.ecref_encode(k, m, rows.ctypes.data_as(ctypes.c_void_p),
+                      _ptr_array(data), _ptr_array(parity),
+                      ctypes.c_size_t(length))
+    return parity
+
+
+def decode(technique, k, m, chunks, present):
+    """chunks: list of k+m uint8 arrays (erased entries are overwritten in
+    place with the reconstruction). present: list/array of 0/1 flags."""
+    t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    length = chunks[0].nbytes
+    pres = np.asarray(present, dtype=np.uint8)
+    r = _ref.ecref_decode(t, k, m, _ptr_array(chunks),
+                          pres.ctypes.data_as(ctypes.c_void_p),
+                          ctypes.c_size_t(length))
+    if r != 0:
+        raise ValueError(f"ecref_decode failed: {r}")
+    return chunks
+
+
+def xor_region(a, b):
+    out = np.zeros_like(a)
+    _ref.ecref_xor_region(a.ctypes.data_as(ctypes.c_void_p),
+                          b.ctypes.data_as(ctypes.c_void_p),
+                          out.ctypes.data_as(ctypes.c_void_p),
+                          ctypes.c_size_t(a.nbytes))
+    return out
+
+
+def region_mul_xor(coeff, delta, parity):
+    _ref.ecref_region_mul_xor(ctypes.c_uint8(coeff),
+                              delta.ctypes.data_as(ctypes.c_void_p),
+                              parity.ctypes.data_as(ctypes.c_void_p),
+                              ctypes.c_size_t(delta.nbytes))
+    return parity
+
+
+def chunk_size(technique, k, stripe_width, w=8):
+    t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    if t == T_RS_VAN_JERASURE:
+        return _ref.ecref_chunk_size_jerasure(k, w, stripe_width)
+    return _ref.ecref_chunk_size_isa(k, stripe_width)
+
+
+# ---- fast CPU-baseline batch ops (AVX2/OpenMP; kind="port" in bench) ----
+
+def cpu_threads():
+    return _cpu.eccpu_threads()
+
+
+def cpu_encode_batch(technique, k, m, batch, n_stripes, chunk_bytes):
+    """batch: uint8 ndarray of n_stripes*(k+m)*chunk_bytes laid out like the
+    GPU device buffer. Encodes parity in place."""
+    t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    r = _cpu.eccpu_encode_batch(t, k, m,
+                                batch.ctypes.data_as(ctypes.c_void_p),
+                                ctypes.c_long(n_stripes),
+                                ctypes.c_size_t(chunk_bytes))
+    if r != 0:
+        raise ValueError(f"eccpu_encode_batch failed: {r}")
+
+
+def cpu_decode_batch(technique, k, m, batch, present, n_stripes, chunk_bytes):
+    t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    pres = np.asarray(present, dtype=np.uint8)
+    r = _cpu.eccpu_decode_batch(t, k, m,
+                                batch.ctypes.data_as(ctypes.c_void_p),
+                                pres.ctypes.data_as(ctypes.c_void_p),
+                                ctypes.c_long(n_stripes),
+                                ctypes.c_size_t(chunk_bytes))
+    if r != 0:
+        raise ValueError(f"eccpu_decode_batch failed: {r}")

@@ -1,0 +1,399 @@
+/* ec_ref.c — CPU oracle. TEST INFRASTRUCTURE ONLY (see ec_ref.h header for
+ * scope, reference citations and the parity-pinning status). */
+#include "ec_ref.h"
+
+#include <errno.h>
+#include <string.h>
+
+/* ---------------- GF(2^8) over 0x11d ---------------- */
+
+static uint8_t gf_log[256];
+static uint8_t gf_exp[256];
+static int gf_ready = 0;
+
+void ecref_gf_init(void)
+{
+  if (gf_ready)
+    return;
+  /* generator 2, primitive polynomial x^8+x^4+x^3+x^2+1 (0x11d): the field
+   * of gf-complete w=8 and isa-l (published; submodules absent, ec_ref.h). */
+  unsigned v = 1;
+  for (int i = 0; i < 255; i++) {
+    gf_exp[i] = (uint8_t)v;
+    gf_log[v] = (uint8_t)i;
+    v <<= 1;
+    if (v & 0x100)
+      v ^= 0x11d;
+  }
+  gf_exp[255] = gf_exp[0]; /* convenience wrap */
+  gf_log[0] = 0;           /* undefined; callers must special-case 0 */
+  gf_ready = 1;
+}
+
+uint8_t ecref_gf_mul(uint8_t a, uint8_t b)
+{
+  if (a == 0 || b == 0)
+    return 0;
+  int s = gf_log[a] + gf_log[b];
+  if (s >= 255)
+    s -= 255;
+  return gf_exp[s];
+}
+
+uint8_t ecref_gf_inv(uint8_t a)
+{
+  if (a == 0)
+    return 0; /* isa-l gf_inv(0) returns 0 */
+  return gf_exp[255 - gf_log[a]];
+}
+
+const uint8_t *ecref_gf_log_table(void) { ecref_gf_init(); return gf_log; }
+const uint8_t *ecref_gf_exp_table(void) { ecref_gf_init(); return gf_exp; }
+
+static uint8_t gf_div(uint8_t a, uint8_t b)
+{
+  if (a == 0)
+    return 0;
+  /* b == 0 is a caller bug; mirror gf arithmetic by returning 0 */
+  if (b == 0)
+    return 0;
+  int s = gf_log[a] - gf_log[b];
+  if (s < 0)
+    s += 255;
+  return gf_exp[s];
+}
+
+/* ---------------- generator matrices ---------------- */
+
+/* Restates isa-l gf_gen_rs_matrix (used at ErasureCodeIsa.cc:659):
+ * identity on top; row k+i = [g^0, g^1, ..., g^(k-1)] with g = 2^i.
+ * The first coding row (i=0, g=1) is all ones => parity0 = XOR of data,
+ * the structural fact the reference's single-erasure fast path relies on
+ * (ErasureCodeIsa.cc:395-456). */
+int ecref_matrix_rs_vandermonde_isa(uint8_t *a, int k, int m)
+{
+  ecref_gf_init();
+  if (k < 1 || m < 0 || k + m > 255)
+    return -EINVAL;
+  memset(a, 0, (size_t)(k + m) * k);
+  for (int i = 0; i < k; i++)
+    a[(size_t)k * i + i] = 1;
+  uint8_t gen = 1;
+  for (int i = k; i < k + m; i++) {
+    uint8_t p = 1;
+    for (int j = 0; j < k; j++) {
+      a[(size_t)k * i + j] = p;
+      p = ecref_gf_mul(p, gen);
+    }
+    gen = ecref_gf_mul(gen, 2);
+  }
+  return 0;
+}
+
+/* Restates isa-l gf_gen_cauchy1_matrix (ErasureCodeIsa.cc:661):
+ * identity on top; coding element (i,j) = 1/(i XOR j), i in [k, k+m). */
+int ecref_matrix_cauchy_isa(uint8_t *a, int k, int m)
+{
+  ecref_gf_init();
+  if (k < 1 || m < 0 || k + m > 255)
+    return -EINVAL;
+  memset(a, 0, (size_t)(k + m) * k);
+  for (int i = 0; i < k; i++)
+    a[(size_t)k * i + i] = 1;
+  uint8_t *p = a + (size_t)k * k;
+  for (int i = k; i < k + m; i++)
+    for (int j = 0; j < k; j++)
+      *p++ = ecref_gf_inv((uint8_t)(i ^ j));
+  return 0;
+}
+
+/* Restates jerasure-2.0 reed_sol.c (submodule absent; published algorithm):
+ * reed_sol_extended_vandermonde_matrix -> systematic reduction ->
+ * first-coding-row normalised to all ones. Used by the reference at
+ * ErasureCodeJerasure.cc:431-435 via reed_sol_vandermonde_coding_matrix.
+ * w=8 only here. Byte-level agreement with compiled jerasure is an
+ * assumption to be spot-verified (ec_ref.h); the structural properties
+ * (systematic top identity, all-ones first coding row, MDS) are test-pinned.
+ */
+int ecref_matrix_rs_vandermonde_jerasure(uint8_t *a, int k, int m)
+{
+  ecref_gf_init();
+  int rows = k + m, cols = k;
+  if (k < 1 || m < 0 || rows > 255)
+    return -EINVAL;
+
+  /* extended Vandermonde: row 0 = e_0; rows 1..rows-2: [i^0, i^1, ...];
+   * last row = e_{cols-1}. */
+  for (int j = 0; j < cols; j++)
+    a[j] = (j == 0);
+  if (rows > 1)
+    for (int j = 0; j < cols; j++)
+      a[(size_t)(rows - 1) * cols + j] = (j == cols - 1);
+  for (int i = 1; i < rows - 1; i++) {
+    uint8_t v = 1;
+    for (int j = 0; j < cols; j++) {
+      a[(size_t)i * cols + j] = v;
+      v = ecref_gf_mul(v, (uint8_t)i);
+    }
+  }
+
+  /* Systematic reduction by column operations (row swap for pivoting),
+   * mirroring reed_sol_big_vandermonde_distribution_matrix. */
+  for (int i = 1; i < cols; i++) {
+    /* pivot: find row j >= i with a[j][i] != 0, swap into row i */
+    int j = i;
+    while (j < rows && a[(size_t)j * cols + i] == 0)
+      j++;
+    if (j >= rows)
+      return -EDOM; /* cannot happen for valid (k,m,w) */
+    if (j != i)
+      for (int c = 0; c < cols; c++) {
+        uint8_t t = a[(size_t)j * cols + c];
+        a[(size_t)j * cols + c] = a[(size_t)i * cols + c];
+        a[(size_t)i * cols + c] = t;
+      }
+    /* scale column i so a[i][i] == 1 */
+    uint8_t piv = a[(size_t)i * cols + i];
+    if (piv != 1) {
+      uint8_t inv = gf_div(1, piv);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + i] = ecref_gf_mul(inv, a[(size_t)r * cols + i]);
+    }
+    /* eliminate row i outside the pivot column: col j ^= a[i][j] * col i */
+    for (int c = 0; c < cols; c++) {
+      uint8_t t = a[(size_t)i * cols + c];
+      if (c != i && t != 0)
+        for (int r = 0; r < rows; r++)
+          a[(size_t)r * cols + c] ^=
+              ecref_gf_mul(t, a[(size_t)r * cols + i]);
+    }
+  }
+
+  /* Normalise the first coding row (row cols) to all ones: scale each
+   * column j by 1/a[cols][j], then restore the identity diagonal by scaling
+   * row j (which holds only the diagonal element) back. */
+  for (int j = 0; j < cols; j++) {
+    uint8_t t = a[(size_t)cols * cols + j];
+    if (t != 0 && t != 1) {
+      uint8_t inv = gf_div(1, t);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + j] = ecref_gf_mul(inv, a[(size_t)r * cols + j]);
+      for (int c = 0; c < cols; c++)
+        a[(size_t)j * cols + c] = ecref_gf_mul(t, a[(size_t)j * cols + c]);
+    }
+  }
+  return 0;
+}
+
+int ecref_matrix(int technique, uint8_t *a, int k, int m)
+{
+  switch (technique) {
+  case ECREF_T_RS_VAN_ISA:      return ecref_matrix_rs_vandermonde_isa(a, k, m);
+  case ECREF_T_CAUCHY_ISA:      return ecref_matrix_cauchy_isa(a, k, m);
+  case ECREF_T_RS_VAN_JERASURE: return ecref_matrix_rs_vandermonde_jerasure(a, k, m);
+  default:                      return -EINVAL;
+  }
+}
+
+/* Restates isa-l gf_invert_matrix (used at ErasureCodeIsa.cc:535):
+ * Gauss-Jordan with row-swap pivoting. in_mat clobbered. */
+int ecref_gf_invert_matrix(uint8_t *in_mat, uint8_t *out_mat, int k)
+{
+  ecref_gf_init();
+  memset(out_mat, 0, (size_t)k * k);
+  for (int i = 0; i < k; i++)
+    out_mat[(size_t)k * i + i] = 1;
+
+  for (int i = 0; i < k; i++) {
+    if (in_mat[(size_t)k * i + i] == 0) {
+      int j = i + 1;
+      while (j < k && in_mat[(size_t)k * j + i] == 0)
+        j++;
+      if (j >= k)
+        return -1; /* singular */
+      for (int c = 0; c < k; c++) {
+        uint8_t t = in_mat[(size_t)k * i + c];
+        in_mat[(size_t)k * i + c] = in_mat[(size_t)k * j + c];
+        in_mat[(size_t)k * j + c] = t;
+        t = out_mat[(size_t)k * i + c];
+        out_mat[(size_t)k * i + c] = out_mat[(size_t)k * j + c];
+        out_mat[(size_t)k * j + c] = t;
+      }
+    }
+    uint8_t piv = in_mat[(size_t)k * i + i];
+    uint8_t inv = ecref_gf_inv(piv);
+    for (int c = 0; c < k; c++) {
+      in_mat[(size_t)k * i + c] = ecref_gf_mul(inv, in_mat[(size_t)k * i + c]);
+      out_mat[(size_t)k * i + c] = ecref_gf_mul(inv, out_mat[(size_t)k * i + c]);
+    }
+    for (int r = 0; r < k; r++) {
+      if (r == i)
+        continue;
+      uint8_t f = in_mat[(size_t)k * r + i];
+      if (f == 0)
+        continue;
+      for (int c = 0; c < k; c++) {
+        in_mat[(size_t)k * r + c] ^= ecref_gf_mul(f, in_mat[(size_t)k * i + c]);
+        out_mat[(size_t)k * r + c] ^= ecref_gf_mul(f, out_mat[(size_t)k * i + c]);
+      }
+    }
+  }
+  return 0;
+}
+
+/* ---------------- region ops ---------------- */
+
+void ecref_encode(int k, int m, const uint8_t *coding_rows,
+                  const uint8_t *const *data, uint8_t *const *parity,
+                  size_t len)
+{
+  ecref_gf_init();
+  for (int j = 0; j < m; j++) {
+    uint8_t *out = parity[j];
+    memset(out, 0, len);
+    for (int i = 0; i < k; i++) {
+      const uint8_t *d = data[i];
+      if (d == NULL)
+        continue; /* zeros chunk: contributes nothing */
+      uint8_t c = coding_rows[(size_t)j * k + i];
+      if (c == 0)
+        continue;
+      if (c == 1) {
+        for (size_t b = 0; b < len; b++)
+          out[b] ^= d[b];
+      } else {
+        const uint8_t lc = gf_log[c];
+        for (size_t b = 0; b < len; b++) {
+          uint8_t v = d[b];
+          if (v) {
+            int s = lc + gf_log[v];
+            if (s >= 255)
+              s -= 255;
+            out[b] ^= gf_exp[s];
+          }
+        }
+      }
+    }
+  }
+}
+
+int ecref_decode(int technique, int k, int m,
+                 uint8_t *const *chunks, const uint8_t *present,
+                 size_t len)
+{
+  ecref_gf_init();
+  uint8_t gen[255 * 255];
+  if (ecref_matrix(technique, gen, k, m) != 0)
+    return -1;
+
+  int n = k + m;
+  int nerrs = 0;
+  int erasures[255];
+  for (int i = 0; i < n; i++)
+    if (!present[i])
+      erasures[nerrs++] = i;
+  if (nerrs == 0)
+    return 0;
+  if (nerrs > m)
+    return -1;
+
+  /* survivor selection: first k present in id order
+   * (ErasureCodeIsa.cc:483-494 decode_index / ErasureCode.cc:154-170). */
+  int decode_index[255];
+  {
+    int r = 0;
+    for (int i = 0; i < k; i++, r++) {
+      while (r < n && !present[r])
+        r++;
+      if (r >= n)
+        return -1;
+      decode_index[i] = r;
+    }
+  }
+
+  /* b = survivor rows of the generator; d = b^-1
+   * (ErasureCodeIsa.cc:518-535). */
+  uint8_t b[255 * 255], d[255 * 255], c[255 * 255];
+  for (int i = 0; i < k; i++)
+    memcpy(&b[(size_t)i * k], &gen[(size_t)decode_index[i] * k], k);
+  if (ecref_gf_invert_matrix(b, d, k) != 0)
+    return -1;
+
+  /* decode rows: data erasure -> row of d; parity erasure -> generator row
+   * composed with d (ErasureCodeIsa.cc:540-557). */
+  for (int p = 0; p < nerrs; p++) {
+    if (erasures[p] < k) {
+      memcpy(&c[(size_t)p * k], &d[(size_t)erasures[p] * k], k);
+    } else {
+      for (int i = 0; i < k; i++) {
+        uint8_t s = 0;
+        for (int j = 0; j < k; j++)
+          s ^= ecref_gf_mul(d[(size_t)j * k + i],
+                            gen[(size_t)erasures[p] * k + j]);
+        c[(size_t)p * k + i] = s;
+      }
+    }
+  }
+
+  const uint8_t *src[255];
+  uint8_t *dst[255];
+  for (int i = 0; i < k; i++)
+    src[i] = chunks[decode_index[i]];
+  for (int p = 0; p < nerrs; p++)
+    dst[p] = chunks[erasures[p]];
+  ecref_encode(k, nerrs, c, src, dst, len);
+  return 0;
+}
+
+void ecref_xor_region(const uint8_t *a, const uint8_t *b, uint8_t *out, size_t len)
+{
+  for (size_t i = 0; i < len; i++)
+    out[i] = a[i] ^ b[i];
+}
+
+void ecref_region_mul_xor(uint8_t coeff, const uint8_t *delta, uint8_t *parity,
+                          size_t len)
+{
+  ecref_gf_init();
+  if (coeff == 0)
+    return;
+  if (coeff == 1) {
+    for (size_t i = 0; i < len; i++)
+      parity[i] ^= delta[i];
+    return;
+  }
+  const uint8_t lc = gf_log[coeff];
+  for (size_t i = 0; i < len; i++) {
+    uint8_t v = delta[i];
+    if (v) {
+      int s = lc + gf_log[v];
+      if (s >= 255)
+        s -= 255;
+      parity[i] ^= gf_exp[s];
+    }
+  }
+}
+
+/* ErasureCodeIsa.cc:65-79: chunk = ceil(width/k) rounded up to 32. */
+unsigned ecref_chunk_size_isa(int k, unsigned stripe_width)
+{
+  unsigned chunk = (stripe_width + k - 1) / k;
+  unsigned mod = chunk % 32u;
+  if (mod)
+    chunk += 32u - mod;
+  return chunk;
+}
+
+/* ErasureCodeJerasure.cc:85-108 (per_chunk_alignment=false default):
+ * pad the stripe to a multiple of get_alignment() = k*w*sizeof(int)
+ * (w*sizeof(int) % 16 == 0 for w=8, so no LARGEST_VECTOR_WORDSIZE bump),
+ * then divide by k. */
+unsigned ecref_chunk_size_jerasure(int k, int w, unsigned stripe_width)
+{
+  unsigned alignment = (unsigned)k * w * 4u;
+  if ((w * 4u) % 16u)
+    alignment = (unsigned)k * w * 16u;
+  unsigned tail = stripe_width % alignment;
+  unsigned padded = stripe_width + (tail ? alignment - tail : 0);
+  return padded / k;
+}
