@@ -1,0 +1,254 @@
+#!/usr/bin/env python3
+"""bench.py — measures BASELINE.json's metric: EC encode+decode GiB/s,
+RS(k=8, m=3), 1 MiB chunks, on 1..8 MI355X.
+
+One step = one encode pass + one decode pass (3 erasures) of the hot path
+over a device-resident 4096-stripe batch per GPU (configs[1] of
+BASELINE.json; decode erasure count = m per configs[2] style). `value` is
+whole-job input-byte throughput with the reference benchmark's accounting
+(seconds per iterations * input bytes, ceph_erasure_code_benchmark.cc:193):
+encode processes k*C*S input bytes, decode processes k*C*S, so one step
+accounts 2*k*C*S per GPU. Inputs are device-resident random bytes
+(seed 0xEC via splitmix64 fill) — BASELINE.md requires non-constant fill.
+
+Multi-GPU: stripes shard; each rank owns its batch end-to-end (weak
+scaling); torch.distributed (RCCL) is used only for barriers and the
+max-over-ranks timing reduction — no data-path collective.
+
+Every measured run is preceded by a bit-exact parity check of sampled
+stripes against the CPU oracle (BASELINE.md "Parity").
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+GIB = 1024 ** 3
+
+
+def splitmix64(idx, seed):
+    """numpy replica of the device fill (ec_core.hip ecx_splitmix64)."""
+    x = (np.uint64(seed) ^ idx.astype(np.uint64)) + np.uint64(0x9E3779B97F4A7C15)
+    with np.errstate(over="ignore"):
+        x = (x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        x = (x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+    return x ^ (x >> np.uint64(31))
+
+
+def expected_fill(byte_off, nbytes, seed):
+    """Bytes the device fill kernel wrote at [byte_off, byte_off+nbytes)."""
+    assert byte_off % 8 == 0 and nbytes % 8 == 0
+    idx = np.arange(byte_off // 8, (byte_off + nbytes) // 8, dtype=np.uint64)
+    return splitmix64(idx, seed).view(np.uint8)
+
+
+def parity_selfcheck(ctx, dptr, args, seed, sample_stripes=2):
+    """Download a few encoded stripes, re-encode with the oracle from the
+    deterministic fill, compare bit-exactly."""
+    import oracle
+    k, m, C, S = args.k, args.m, args.chunk_bytes, args.stripes
+    n = k + m
+    stripe_bytes = n * C
+    rng = np.random.default_rng(123)
+    for s in sorted(rng.choice(S, size=min(sample_stripes, S), replace=False)):
+        host = np.zeros(stripe_bytes, dtype=np.uint8)
+        # download stripe s
+        import ctypes
+        import ceph_amd
+        src = ctypes.c_void_p(dptr.value + int(s) * stripe_bytes)
+        ctx.download(host, src)
+        data = [host[i * C:(i + 1) * C] for i in range(k)]
+        # data region must equal the deterministic fill
+        exp = expected_fill(int(s) * stripe_bytes, k * C, seed)
+        if not np.array_equal(host[:k * C], exp):
+            raise AssertionError(f"stripe {s}: data region != expected fill")
+        want = oracle.encode(args.technique, k, m, data)
+        for j in range(m):
+            got = host[(k + j) * C:(k + j + 1) * C]
+            if not np.array_equal(got, want[j]):
+                raise AssertionError(
+                    f"stripe {s}: GPU parity {j} mismatches oracle")
+
+
+def cpu_baseline(args, budget_s=12.0):
+    """Time the oracle's ISA-L-class AVX2/OpenMP path (kind='port') on a
+    bounded sample of the same workload on this host's cores."""
+    import oracle
+    k, m, C = args.k, args.m, args.chunk_bytes
+    S = max(1, min(args.stripes, int(2 * GIB / ((k + m) * C))))  # <=2 GiB data
+    batch = np.empty(S * (k + m) * C, dtype=np.uint8)
+    batch[:] = np.frombuffer(os.urandom(1 << 20), np.uint8).repeat(
+        (batch.nbytes + (1 << 20) - 1) // (1 << 20))[:batch.nbytes]
+    present = np.ones(k + m, np.uint8)
+    present[sorted(np.random.default_rng(0xEC).choice(
+        k + m, size=args.erasures, replace=False))] = 0
+    # warm
+    oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)
+    t0 = time.perf_counter()
+    iters = 0
+    while time.perf_counter() - t0 < budget_s:
+        oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)
+        oracle.cpu_decode_batch(args.technique, k, m, batch, present, S, C)
+        iters += 1
+    dt = time.perf_counter() - t0
+    gibs = iters * 2 * k * C * S / GIB / dt
+    return {
+        "value": round(gibs, 3),
+        "unit": "GiB/s",
+        "cores": oracle.cpu_threads(),
+        "kind": "port",
+        "sample": (f"{iters}x encode+decode of a {S}-stripe batch "
+                   f"(k={k},m={m},C={C}) in {dt:.1f}s on host cores"),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--stripes", type=int, default=4096,
+                    help="stripes per GPU (weak scaling)")
+    ap.add_argument("--chunk-kib", type=int, default=1024)
+    ap.add_argument("--k", type=int, default=8)
+    ap.add_argument("--m", type=int, default=3)
+    ap.add_argument("--technique", default="reed_sol_van")
+    ap.add_argument("--erasures", type=int, default=3)
+    ap.add_argument("--seed", type=lambda x: int(x, 0), default=0xEC)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--no-selfcheck", action="store_true")
+    ap.add_argument("--streams", type=int, default=2)
+    args = ap.parse_args()
+    args.chunk_bytes = args.chunk_kib * 1024
+
+    import ceph_amd
+
+    if ceph_amd.device_count() < 1:
+        print(json.dumps({"error": "no GPU visible; bench requires MI355X"}))
+        sys.exit(1)
+
+    # distributed setup (torchrun provides RANK/WORLD_SIZE/LOCAL_RANK)
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        torch.cuda.set_device(local_rank)
+        tdist.init_process_group("nccl")
+        dist = tdist
+
+    k, m, C, S = args.k, args.m, args.chunk_bytes, args.stripes
+    n = k + m
+    buf_bytes = S * n * C
+    seed = args.seed + rank
+
+    ctx = ceph_amd.EcContext(k, m, args.technique, device=local_rank,
+                             n_streams=args.streams)
+    dptr = ctx.dbuf_alloc(buf_bytes)
+    ctx.fill_random(dptr, buf_bytes, seed)
+    ctx.sync()
+
+    present_mask = (1 << n) - 1
+    erased = sorted(np.random.default_rng(args.seed).choice(
+        n, size=args.erasures, replace=False).tolist())
+    for e in erased:
+        present_mask &= ~(1 << e)
+
+    def step(timed_accum=None):
+        ctx.encode_batch(dptr, S, C)
+        if timed_accum is not None:
+            timed_accum.append(ctx.last_kernel_ms())
+        ctx.decode_batch(dptr, S, C, present_mask)
+        ctx.sync()
+
+    # parity self-check before measuring (rank 0)
+    step()
+    if rank == 0 and not args.no_selfcheck:
+        parity_selfcheck(ctx, dptr, args, seed)
+
+    for _ in range(args.warmup):
+        step()
+    if dist:
+        dist.barrier()
+        import torch
+        torch.cuda.synchronize()
+
+    enc_ms = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step(enc_ms)
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+        torch.cuda.synchronize()
+        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        dist.barrier()
+
+    if rank == 0:
+        input_bytes_per_step = 2 * k * C * S  # encode + decode accounting
+        value = world * input_bytes_per_step * args.steps / GIB / elapsed
+        enc_kernel_ms = float(np.mean(enc_ms)) if enc_ms else None
+        # encode kernel algorithmic traffic: read k*C*S + write m*C*S
+        alg_bytes = (k + m) * C * S
+        achieved = alg_bytes / (enc_kernel_ms * 1e-3) / 1e9 if enc_kernel_ms else None
+        peak = 8000.0  # GB/s, MI355X HBM3E spec (MI355X_MICROARCH.md)
+        line = {
+            "metric": "EC encode+decode GiB/s",
+            "value": round(value, 2),
+            "unit": "GiB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no absolute EC numbers
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": ("RS-Vandermonde k=8 m=3, 1 MiB chunks, "
+                             "4096-stripe batch encode+decode, "
+                             "device-resident (BASELINE configs[1]+[2])"),
+                "k": k, "m": m, "chunk_bytes": C, "stripes_per_gpu": S,
+                "technique": args.technique, "erasures": erased,
+                "seed": hex(args.seed),
+                "encode_gibs": round(world * k * C * S * args.steps / GIB /
+                                     elapsed * 2, 2) if False else None,
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": round(achieved, 1) if achieved else None,
+                "peak": peak,
+                "unit": "GB/s",
+                "frac": round(achieved / peak, 4) if achieved else None,
+                "traffic": None,  # PMC traffic comes from rocprofv3 runs
+                                  # committed under profiles/
+                "kernel": "ec_gf_matmul_kernel<3,false> (encode)",
+                "kernel_ms": round(enc_kernel_ms, 4) if enc_kernel_ms else None,
+                "alg_bytes_per_launch": alg_bytes,
+            },
+            "cpu_baseline": (cpu_baseline(args)
+                             if (world == 1 and not args.no_cpu_baseline)
+                             else None),
+        }
+        del line["config"]["encode_gibs"]
+        print(json.dumps(line))
+
+    ctx.dbuf_free(dptr)
+    ctx.close()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,191 @@
+// gf.cpp — see gf.h. Product-side host GF(2^8) + matrix derivation.
+#include "gf.h"
+
+#include <cstring>
+
+namespace ecx {
+
+GF8::GF8() {
+  unsigned v = 1;
+  for (int i = 0; i < 255; i++) {
+    exp[i] = (uint8_t)v;
+    log[v] = (uint8_t)i;
+    v <<= 1;
+    if (v & 0x100) v ^= 0x11d;
+  }
+  exp[255] = exp[0];
+  log[0] = 0;
+}
+
+const GF8 &gf8() {
+  static const GF8 f;
+  return f;
+}
+
+// isa-l gf_gen_rs_matrix (ErasureCodeIsa.cc:659): identity ++ power rows,
+// row k+i = [(2^i)^j], so the first coding row is all ones.
+bool gen_matrix_rs_van_isa(std::vector<uint8_t> &a, int k, int m) {
+  const GF8 &f = gf8();
+  if (k < 1 || m < 0 || k + m > 255) return false;
+  a.assign((size_t)(k + m) * k, 0);
+  for (int i = 0; i < k; i++) a[(size_t)k * i + i] = 1;
+  uint8_t gen = 1;
+  for (int i = k; i < k + m; i++) {
+    uint8_t p = 1;
+    for (int j = 0; j < k; j++) {
+      a[(size_t)k * i + j] = p;
+      p = f.mul(p, gen);
+    }
+    gen = f.mul(gen, 2);
+  }
+  return true;
+}
+
+// isa-l gf_gen_cauchy1_matrix (ErasureCodeIsa.cc:661).
+bool gen_matrix_cauchy_isa(std::vector<uint8_t> &a, int k, int m) {
+  const GF8 &f = gf8();
+  if (k < 1 || m < 0 || k + m > 255) return false;
+  a.assign((size_t)(k + m) * k, 0);
+  for (int i = 0; i < k; i++) a[(size_t)k * i + i] = 1;
+  for (int i = k; i < k + m; i++)
+    for (int j = 0; j < k; j++) a[(size_t)k * i + j] = f.inv((uint8_t)(i ^ j));
+  return true;
+}
+
+// jerasure reed_sol_vandermonde_coding_matrix, w=8 (published jerasure-2.0
+// algorithm; used by ErasureCodeJerasure.cc:431-435). Extended Vandermonde
+// -> systematic reduction by column ops -> first coding row normalised to
+// all ones.
+bool gen_matrix_rs_van_jerasure(std::vector<uint8_t> &a, int k, int m) {
+  const GF8 &f = gf8();
+  int rows = k + m, cols = k;
+  if (k < 1 || m < 0 || rows > 255) return false;
+  a.assign((size_t)rows * cols, 0);
+  a[0] = 1;
+  if (rows > 1) a[(size_t)(rows - 1) * cols + (cols - 1)] = 1;
+  for (int i = 1; i < rows - 1; i++) {
+    uint8_t v = 1;
+    for (int j = 0; j < cols; j++) {
+      a[(size_t)i * cols + j] = v;
+      v = f.mul(v, (uint8_t)i);
+    }
+  }
+  for (int i = 1; i < cols; i++) {
+    int j = i;
+    while (j < rows && a[(size_t)j * cols + i] == 0) j++;
+    if (j >= rows) return false;
+    if (j != i)
+      for (int c = 0; c < cols; c++)
+        std::swap(a[(size_t)j * cols + c], a[(size_t)i * cols + c]);
+    uint8_t piv = a[(size_t)i * cols + i];
+    if (piv != 1) {
+      uint8_t inv = f.div(1, piv);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + i] = f.mul(inv, a[(size_t)r * cols + i]);
+    }
+    for (int c = 0; c < cols; c++) {
+      uint8_t t = a[(size_t)i * cols + c];
+      if (c != i && t != 0)
+        for (int r = 0; r < rows; r++)
+          a[(size_t)r * cols + c] ^= f.mul(t, a[(size_t)r * cols + i]);
+    }
+  }
+  for (int j = 0; j < cols; j++) {
+    uint8_t t = a[(size_t)cols * cols + j];
+    if (t != 0 && t != 1) {
+      uint8_t inv = f.div(1, t);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + j] = f.mul(inv, a[(size_t)r * cols + j]);
+      for (int c = 0; c < cols; c++)
+        a[(size_t)j * cols + c] = f.mul(t, a[(size_t)j * cols + c]);
+    }
+  }
+  return true;
+}
+
+bool gen_matrix(int technique, std::vector<uint8_t> &a, int k, int m) {
+  switch (technique) {
+    case 0: return gen_matrix_rs_van_isa(a, k, m);
+    case 1: return gen_matrix_cauchy_isa(a, k, m);
+    case 2: return gen_matrix_rs_van_jerasure(a, k, m);
+    default: return false;
+  }
+}
+
+bool gf_invert(const uint8_t *in, uint8_t *out, int k) {
+  const GF8 &f = gf8();
+  std::vector<uint8_t> w(in, in + (size_t)k * k);
+  std::memset(out, 0, (size_t)k * k);
+  for (int i = 0; i < k; i++) out[(size_t)k * i + i] = 1;
+  for (int i = 0; i < k; i++) {
+    if (w[(size_t)k * i + i] == 0) {
+      int j = i + 1;
+      while (j < k && w[(size_t)k * j + i] == 0) j++;
+      if (j >= k) return false;
+      for (int c = 0; c < k; c++) {
+        std::swap(w[(size_t)k * i + c], w[(size_t)k * j + c]);
+        std::swap(out[(size_t)k * i + c], out[(size_t)k * j + c]);
+      }
+    }
+    uint8_t inv = f.inv(w[(size_t)k * i + i]);
+    for (int c = 0; c < k; c++) {
+      w[(size_t)k * i + c] = f.mul(inv, w[(size_t)k * i + c]);
+      out[(size_t)k * i + c] = f.mul(inv, out[(size_t)k * i + c]);
+    }
+    for (int r = 0; r < k; r++) {
+      if (r == i) continue;
+      uint8_t t = w[(size_t)k * r + i];
+      if (!t) continue;
+      for (int c = 0; c < k; c++) {
+        w[(size_t)k * r + c] ^= f.mul(t, w[(size_t)k * i + c]);
+        out[(size_t)k * r + c] ^= f.mul(t, out[(size_t)k * i + c]);
+      }
+    }
+  }
+  return true;
+}
+
+bool compose_decode_rows(const std::vector<uint8_t> &gen, int k, int m,
+                         uint64_t present_mask,
+                         std::vector<int> &survivors,
+                         std::vector<int> &erased,
+                         std::vector<uint8_t> &rows) {
+  const GF8 &f = gf8();
+  int n = k + m;
+  survivors.clear();
+  erased.clear();
+  for (int i = 0; i < n; i++) {
+    if (present_mask & (1ull << i)) {
+      if ((int)survivors.size() < k) survivors.push_back(i);
+    } else {
+      erased.push_back(i);
+    }
+  }
+  if ((int)survivors.size() < k) return false;
+  if ((int)erased.size() > m) return false;
+
+  std::vector<uint8_t> b((size_t)k * k), d((size_t)k * k);
+  for (int i = 0; i < k; i++)
+    std::memcpy(&b[(size_t)i * k], &gen[(size_t)survivors[i] * k], k);
+  if (!gf_invert(b.data(), d.data(), k)) return false;
+
+  rows.assign(erased.size() * (size_t)k, 0);
+  for (size_t p = 0; p < erased.size(); p++) {
+    int e = erased[p];
+    if (e < k) {
+      std::memcpy(&rows[p * k], &d[(size_t)e * k], k);
+    } else {
+      // lost parity: compose generator row with the inverse
+      // (ErasureCodeIsa.cc:546-557)
+      for (int i = 0; i < k; i++) {
+        uint8_t s = 0;
+        for (int j = 0; j < k; j++)
+          s ^= f.mul(d[(size_t)j * k + i], gen[(size_t)e * k + j]);
+        rows[p * k + i] = s;
+      }
+    }
+  }
+  return true;
+}
+
+}  // namespace ecx

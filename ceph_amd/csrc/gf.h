@@ -1,0 +1,61 @@
+// gf.h — product-side GF(2^8) host arithmetic for libec_mi355x_core.
+//
+// Host work on the EC hot path is integers only: generator-matrix derivation
+// once per (k,m,technique) (mirrors prepare(), src/erasure-code/isa/
+// ErasureCodeIsa.cc:637-697 and src/erasure-code/jerasure/
+// ErasureCodeJerasure.cc:431-435) and per-erasure-signature decode-row
+// composition (ErasureCodeIsa.cc:510-567). The byte/stripe work itself is on
+// the GPU (ec_core.hip). Independent implementation from oracle/ec_ref.c —
+// the oracle is the checker, this is the product.
+#pragma once
+
+#include <cstdint>
+#include <vector>
+
+namespace ecx {
+
+// GF(2^8) over the primitive polynomial 0x11d (the field of gf-complete w=8
+// and isa-l).
+struct GF8 {
+  uint8_t log[256];
+  uint8_t exp[256];
+  GF8();
+  uint8_t mul(uint8_t a, uint8_t b) const {
+    if (!a || !b) return 0;
+    int s = log[a] + log[b];
+    if (s >= 255) s -= 255;
+    return exp[s];
+  }
+  uint8_t inv(uint8_t a) const { return a ? exp[255 - log[a]] : 0; }
+  uint8_t div(uint8_t a, uint8_t b) const {
+    if (!a || !b) return 0;
+    int s = log[a] - log[b];
+    if (s < 0) s += 255;
+    return exp[s];
+  }
+};
+const GF8 &gf8();
+
+// Full (k+m) x k generator, identity on top, row-major (isa-l layout).
+// Returns false on invalid parameters.
+bool gen_matrix_rs_van_isa(std::vector<uint8_t> &a, int k, int m);
+bool gen_matrix_cauchy_isa(std::vector<uint8_t> &a, int k, int m);
+bool gen_matrix_rs_van_jerasure(std::vector<uint8_t> &a, int k, int m);
+bool gen_matrix(int technique, std::vector<uint8_t> &a, int k, int m);
+
+// Gauss-Jordan inverse of a k x k matrix; false if singular.
+bool gf_invert(const uint8_t *in, uint8_t *out, int k);
+
+// Decode-row composition for a given erasure pattern (mirrors
+// ErasureCodeIsa.cc:510-567): picks the first k present chunks in id order
+// as survivors (ErasureCode.cc:154-170), inverts the survivor submatrix,
+// and emits one coefficient row per erased chunk over those survivors.
+// Outputs: survivors[k] (chunk ids), erased[] (chunk ids, ascending),
+// rows (n_erased x k coefficients). Returns false if undecodable.
+bool compose_decode_rows(const std::vector<uint8_t> &gen, int k, int m,
+                         uint64_t present_mask,
+                         std::vector<int> &survivors,
+                         std::vector<int> &erased,
+                         std::vector<uint8_t> &rows);
+
+}  // namespace ecx

@@ -101,8 +101,8 @@ def encode(technique, k, m, data, chunk_bytes=None):
     """data: list of k uint8 arrays (or None for zeros). Returns list of m
     parity arrays. Scalar oracle (ecref_encode)."""
     lens = {d.nbytes for d in data if d is not None}
-    assert len(lens) == 1
-    length = lens.pop()
+    assert len(lens) == 1 or (not lens and chunk_bytes)
+    length = lens.pop() if lens else chunk_bytes
     gen = matrix(technique, k, m)
     rows = np.ascontiguousarray(gen[k:])
     parity = [np.zeros(length, dtype=np.uint8) for _ in range(m)]

@@ -1,0 +1,209 @@
+"""GPU parity tests: the HIP path must match the CPU oracle bit-exactly on
+the same inputs (BASELINE.md 'Parity'). All tests here need
+an MI355X and run through the C-ABI (include/ec_mi355x.h)."""
+import ctypes
+from itertools import combinations
+
+import numpy as np
+import pytest
+
+import ceph_amd
+import oracle
+
+pytestmark = pytest.mark.gpu
+
+TECHS = ["reed_sol_van", "cauchy", "jerasure_reed_sol_van"]
+
+
+def make_ctx(k, m, tech):
+    return ceph_amd.EcContext(k, m, tech, device=0)
+
+
+@pytest.mark.parametrize("tech", TECHS)
+@pytest.mark.parametrize("k,m", [(2, 1), (4, 2), (8, 3), (10, 4), (12, 4)])
+def test_encode_parity_vs_oracle(tech, k, m):
+    C = 64 * 1024
+    rng = np.random.default_rng(0xEC ^ (k << 8) ^ m)
+    ctx = make_ctx(k, m, tech)
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        got = ctx.encode_chunks(data)
+        want = oracle.encode(tech, k, m, data)
+        for j in range(m):
+            assert np.array_equal(got[j], want[j]), (tech, k, m, j)
+    finally:
+        ctx.close()
+
+
+@pytest.mark.parametrize("tech", TECHS)
+def test_decode_exhaustive_patterns(tech):
+    """Every erasure pattern for (k=4, m=3), mirroring the reference's
+    exhaustive sweeps (TestErasureCodeIsa.cc:400-650)."""
+    k, m = 4, 3
+    C = 16 * 1024
+    rng = np.random.default_rng(5)
+    ctx = make_ctx(k, m, tech)
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        par = ctx.encode_chunks(data)
+        full = data + par
+        for e in range(1, m + 1):
+            for er in combinations(range(k + m), e):
+                present = [i not in er for i in range(k + m)]
+                chunks = [c.copy() if present[i] else np.zeros(C, np.uint8)
+                          for i, c in enumerate(full)]
+                ctx.decode_chunks(chunks, present)
+                for i in range(k + m):
+                    assert np.array_equal(chunks[i], full[i]), (tech, er, i)
+    finally:
+        ctx.close()
+
+
+def test_zero_chunk_convention():
+    """NULL data pointer == zeros chunk (ErasureCodeJerasure.cc:146-157)."""
+    k, m = 6, 3
+    C = 4096
+    rng = np.random.default_rng(9)
+    ctx = make_ctx(k, m, "reed_sol_van")
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        data[2] = None
+        data[5] = None
+        got = ctx.encode_chunks(data)
+        want = oracle.encode("reed_sol_van", k, m, data)
+        for j in range(m):
+            assert np.array_equal(got[j], want[j])
+        # all-None => all-zero parity (zero-in-zero-out flag behaviour)
+        got = ctx.encode_chunks([None] * (k - 1) +
+                                [np.zeros(C, np.uint8)])
+        for j in range(m):
+            assert not got[j].any()
+    finally:
+        ctx.close()
+
+
+@pytest.mark.parametrize("tech", TECHS)
+def test_batch_api_vs_oracle(tech):
+    """Device-resident batch encode+decode (the bench hot path) vs the
+    oracle's batch path on identical bytes."""
+    k, m = 8, 3
+    S, C = 16, 64 * 1024
+    n = k + m
+    ctx = make_ctx(k, m, tech)
+    try:
+        nbytes = S * n * C
+        host = np.random.default_rng(0xEC).integers(
+            0, 256, nbytes, dtype=np.uint8)
+        ref = host.copy()
+        oracle.cpu_encode_batch(tech, k, m, ref, S, C)
+
+        d = ctx.dbuf_alloc(nbytes)
+        ctx.upload(d, host)
+        ctx.encode_batch(d, S, C)
+        out = np.zeros(nbytes, np.uint8)
+        ctx.download(out, d)
+        assert np.array_equal(out, ref), "batch encode mismatch"
+
+        # decode: wipe 3 chunks per stripe on device by re-uploading zeros
+        erased = [1, 6, 9]
+        present_mask = (1 << n) - 1
+        zero = np.zeros(C, np.uint8)
+        for e in erased:
+            present_mask &= ~(1 << e)
+            for s in range(S):
+                dst = ctypes.c_void_p(d.value + (s * n + e) * C)
+                ctx.upload(dst, zero)
+        ctx.decode_batch(d, S, C, present_mask)
+        ctx.download(out, d)
+        assert np.array_equal(out, ref), "batch decode mismatch"
+    finally:
+        ctx.close()
+
+
+def test_fill_random_matches_host_replica():
+    """bench.py's parity self-check depends on reproducing the device fill
+    on the host."""
+    import bench
+    ctx = make_ctx(8, 3, "reed_sol_van")
+    try:
+        nbytes = 1 << 20
+        d = ctx.dbuf_alloc(nbytes)
+        ctx.fill_random(d, nbytes, 0xEC)
+        ctx.sync()
+        out = np.zeros(nbytes, np.uint8)
+        ctx.download(out, d)
+        assert np.array_equal(out, bench.expected_fill(0, nbytes, 0xEC))
+        ctx.dbuf_free(d)
+    finally:
+        ctx.close()
+
+
+def test_delta_ops_vs_oracle():
+    """encode_delta/apply_delta == full re-encode (ParityDelta conformance,
+    TestErasureCodePlugins.cc:302-...)."""
+    k, m = 5, 3
+    C = 16 * 1024
+    tech = "reed_sol_van"
+    rng = np.random.default_rng(21)
+    ctx = make_ctx(k, m, tech)
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        par = ctx.encode_chunks(data)
+        newc = rng.integers(0, 256, C, dtype=np.uint8)
+        delta = ctx.encode_delta(data[2], newc)
+        assert np.array_equal(delta, data[2] ^ newc)
+        par2 = [p.copy() for p in par]
+        for j in range(m):
+            ctx.apply_delta(delta, 2, k + j, par2[j])
+        data[2] = newc
+        want = oracle.encode(tech, k, m, data)
+        for j in range(m):
+            assert np.array_equal(par2[j], want[j]), j
+    finally:
+        ctx.close()
+
+
+def test_large_chunk_and_odd_sizes():
+    """1 MiB chunks (BASELINE shape) and a few non-power-of-2 sizes
+    (multiples of 16 per the C-ABI contract)."""
+    k, m = 8, 3
+    tech = "reed_sol_van"
+    rng = np.random.default_rng(2)
+    ctx = make_ctx(k, m, tech)
+    try:
+        for C in (1 << 20, 4096 + 16, 31 * 16, 1 << 16):
+            data = [rng.integers(0, 256, C, dtype=np.uint8)
+                    for _ in range(k)]
+            got = ctx.encode_chunks(data)
+            want = oracle.encode(tech, k, m, data)
+            for j in range(m):
+                assert np.array_equal(got[j], want[j]), C
+        # non-multiple-of-16 must be rejected loudly, not silently wrong
+        bad = [rng.integers(0, 256, 24, dtype=np.uint8) for _ in range(k)]
+        with pytest.raises(ceph_amd.EcError):
+            ctx.encode_chunks(bad)
+    finally:
+        ctx.close()
+
+
+def test_decode_lru_reuse():
+    """Repeated decode with the same erasure signature must hit the cached
+    plan and stay bit-exact (ErasureCodeIsaTableCache analogue)."""
+    k, m = 8, 3
+    C = 4096
+    ctx = make_ctx(k, m, "reed_sol_van")
+    rng = np.random.default_rng(4)
+    try:
+        for trial in range(5):
+            data = [rng.integers(0, 256, C, dtype=np.uint8)
+                    for _ in range(k)]
+            par = ctx.encode_chunks(data)
+            full = data + par
+            present = [i not in (0, 4, 10) for i in range(k + m)]
+            chunks = [c.copy() if present[i] else np.zeros(C, np.uint8)
+                      for i, c in enumerate(full)]
+            ctx.decode_chunks(chunks, present)
+            for i in range(k + m):
+                assert np.array_equal(chunks[i], full[i]), (trial, i)
+    finally:
+        ctx.close()

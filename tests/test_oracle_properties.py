@@ -1,0 +1,174 @@
+"""Property suite pinning the oracle, mirroring the reference's own test
+strategy: round-trips, systematic prefix, exhaustive erasure
+sweeps, zero-in-zero-out, parity-delta equivalence, RAID6 XOR facts."""
+from itertools import combinations
+
+import numpy as np
+import pytest
+
+import oracle
+
+TECHS = ["reed_sol_van", "cauchy", "jerasure_reed_sol_van"]
+
+
+@pytest.mark.parametrize("tech", TECHS)
+@pytest.mark.parametrize("k,m", [(2, 1), (3, 2), (8, 3), (12, 4)])
+def test_round_trip_random_erasures(tech, k, m):
+    """TestErasureCodeJerasure.cc:59-132 style round-trip + degraded."""
+    rng = np.random.default_rng(0xEC ^ (k * 16 + m))
+    C = 512
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    par = oracle.encode(tech, k, m, data)
+    full = [d.copy() for d in data] + [p.copy() for p in par]
+    for _ in range(30):
+        e = rng.integers(1, m + 1)
+        er = rng.choice(k + m, size=e, replace=False)
+        present = np.ones(k + m, np.uint8)
+        present[er] = 0
+        test = [c.copy() if present[i] else np.zeros(C, np.uint8)
+                for i, c in enumerate(full)]
+        oracle.decode(tech, k, m, test, present)
+        for i in range(k + m):
+            assert (test[i] == full[i]).all()
+
+
+@pytest.mark.parametrize("tech", TECHS)
+def test_exhaustive_erasures(tech):
+    """Exhaustive all-pattern sweep for (k=4, m=3) — the style of
+    TestErasureCodeIsa.cc:400-650 / benchmark --erasures-generation
+    exhaustive."""
+    k, m = 4, 3
+    rng = np.random.default_rng(7)
+    C = 128
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    par = oracle.encode(tech, k, m, data)
+    full = [d.copy() for d in data] + [p.copy() for p in par]
+    for e in range(1, m + 1):
+        for er in combinations(range(k + m), e):
+            present = np.ones(k + m, np.uint8)
+            present[list(er)] = 0
+            test = [c.copy() if present[i] else np.zeros(C, np.uint8)
+                    for i, c in enumerate(full)]
+            oracle.decode(tech, k, m, test, present)
+            for i in range(k + m):
+                assert (test[i] == full[i]).all(), (er, i)
+
+
+@pytest.mark.parametrize("tech", TECHS)
+def test_zero_in_zero_out(tech):
+    """FLAG_EC_PLUGIN_ZERO_INPUT_ZERO_OUTPUT (TestErasureCodePlugins.cc
+    ZeroInZeroOut): all-zero data => all-zero parity."""
+    k, m = 6, 3
+    zero = [np.zeros(256, np.uint8) for _ in range(k)]
+    for p in oracle.encode(tech, k, m, zero):
+        assert not p.any()
+    # NULL chunks mean zeros (zeros-buffer convention)
+    for p in oracle.encode(tech, k, m, [None] * k, chunk_bytes=256):
+        assert not p.any()
+
+
+def test_systematic_prefix():
+    """Systematic codes: data chunks are the input verbatim
+    (TestErasureCodeJerasure.cc:93-96 memcmp)."""
+    # encode() in the oracle takes chunks directly, so systematicity is the
+    # identity-top property of the matrix — verified in test_gf_kat; here we
+    # check the mixed null/dense convention instead.
+    k, m = 4, 2
+    rng = np.random.default_rng(3)
+    C = 128
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    ref = oracle.encode("reed_sol_van", k, m, data)
+    # zeroing chunk 2 == passing None for chunk 2
+    data2 = [data[0], data[1], None, data[3]]
+    dataz = [data[0], data[1], np.zeros(C, np.uint8), data[3]]
+    a = oracle.encode("reed_sol_van", k, m, data2)
+    b = oracle.encode("reed_sol_van", k, m, dataz)
+    for x, y in zip(a, b):
+        assert (x == y).all()
+    assert not all((x == y).all() for x, y in zip(a, ref))
+
+
+@pytest.mark.parametrize("tech", TECHS)
+def test_parity_delta_equivalence(tech):
+    """ParityDelta conformance (TestErasureCodePlugins.cc:302-...):
+    applying encode_delta + apply_delta for a changed data chunk must equal
+    a full re-encode."""
+    k, m = 5, 3
+    rng = np.random.default_rng(11)
+    C = 256
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    par = oracle.encode(tech, k, m, data)
+    # change data chunk 2
+    newc = rng.integers(0, 256, C, dtype=np.uint8)
+    delta = oracle.xor_region(data[2], newc)
+    g = oracle.matrix(tech, k, m)
+    par2 = [p.copy() for p in par]
+    for j in range(m):
+        oracle.region_mul_xor(int(g[k + j, 2]), delta, par2[j])
+    data[2] = newc
+    want = oracle.encode(tech, k, m, data)
+    for j in range(m):
+        assert (par2[j] == want[j]).all(), (tech, j)
+
+
+def test_raid6_m1_xor():
+    """m=1 parity is the XOR of data for every technique whose first coding
+    row is all ones; isa m==1 always uses plain XOR
+    (ErasureCodeIsa.cc:294-296)."""
+    k = 6
+    rng = np.random.default_rng(5)
+    C = 192
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    x = np.zeros(C, np.uint8)
+    for d in data:
+        x ^= d
+    for tech in ("reed_sol_van", "jerasure_reed_sol_van"):
+        p = oracle.encode(tech, k, 1, data)
+        assert (p[0] == x).all()
+
+
+def test_chunk_size_rules():
+    # isa: ceil(width/k) rounded to 32 (ErasureCodeIsa.cc:65-79)
+    assert oracle.chunk_size("reed_sol_van", 8, 8 * 1024 * 1024) == 1024 * 1024
+    assert oracle.chunk_size("reed_sol_van", 8, 100) == 32
+    assert oracle.chunk_size("reed_sol_van", 7, 4096) == 608
+    # jerasure: stripe padded to k*w*4 then /k (ErasureCodeJerasure.cc:85-108)
+    assert oracle.chunk_size("jerasure_reed_sol_van", 2, 8192) == 4096
+    # k=3: alignment = 3*8*4 = 96; 100 pads to 192; 192/3 = 64
+    assert oracle.chunk_size("jerasure_reed_sol_van", 3, 100) == 64
+
+
+@pytest.mark.parametrize("tech", TECHS)
+def test_cpu_batch_matches_scalar_oracle(tech):
+    """The AVX2/OpenMP baseline must agree bit-exactly with the scalar
+    restatement (it is the bench's cpu_baseline leg)."""
+    k, m = 8, 3
+    S, C = 3, 4096
+    rng = np.random.default_rng(0xA)
+    batch = rng.integers(0, 256, S * (k + m) * C, dtype=np.uint8)
+    oracle.cpu_encode_batch(tech, k, m, batch, S, C)
+    for s in range(S):
+        st = batch[s * (k + m) * C:(s + 1) * (k + m) * C]
+        data = [st[i * C:(i + 1) * C].copy() for i in range(k)]
+        par = oracle.encode(tech, k, m, data)
+        for j in range(m):
+            assert (st[(k + j) * C:(k + j + 1) * C] == par[j]).all()
+    # decode batch round-trip
+    ref = batch.copy()
+    present = np.ones(k + m, np.uint8)
+    present[[0, 5, 9]] = 0
+    for s in range(S):
+        for e in (0, 5, 9):
+            batch[(s * (k + m) + e) * C:(s * (k + m) + e + 1) * C] = 0
+    oracle.cpu_decode_batch(tech, k, m, batch, present, S, C)
+    assert (batch == ref).all()
+
+
+def test_decode_too_many_erasures_fails():
+    k, m = 4, 2
+    C = 64
+    chunks = [np.zeros(C, np.uint8) for _ in range(k + m)]
+    present = np.ones(k + m, np.uint8)
+    present[[0, 1, 2]] = 0  # 3 > m erasures
+    with pytest.raises(ValueError):
+        oracle.decode("reed_sol_van", k, m, chunks, present)
