@@ -1,0 +1,194 @@
+// plugin_mi355x.cc — libec_mi355x.so: the MI355X EC plugin behind the
+// reference's plugin contract. Pattern mirrors the reference's plugin shims
+// (src/erasure-code/jerasure/ErasureCodePluginJerasure.cc:35-90 factory
+// dispatch + entry points; src/erasure-code/isa glue marshalling,
+// ErasureCodeIsa.cc:118-243). All byte/stripe compute goes through the
+// C-ABI into the gfx950 kernels (include/ec_mi355x.h); there is no CPU
+// fallback — init() fails with -ENODEV when no GPU is visible.
+#include <cerrno>
+#include <cstring>
+#include <memory>
+#include <ostream>
+#include <sstream>
+
+#include "../../include/ec_mi355x.h"
+#include "erasure_code_plugin.h"
+
+using namespace ecx;
+
+namespace {
+
+int technique_id(const std::string &t) {
+  if (t == "reed_sol_van") return ECX_T_RS_VAN_ISA;
+  if (t == "cauchy") return ECX_T_CAUCHY_ISA;
+  if (t == "jerasure_reed_sol_van") return ECX_T_RS_VAN_JERASURE;
+  return -1;
+}
+
+class ErasureCodeMi355x final : public ErasureCode {
+  ecx_ctx *ctx_ = nullptr;
+  int k_ = 0, m_ = 0, w_ = 8, device_ = 0, streams_ = 2;
+  std::string technique_;
+
+ public:
+  explicit ErasureCodeMi355x(std::string technique)
+      : technique_(std::move(technique)) {}
+  ~ErasureCodeMi355x() override {
+    if (ctx_) ecx_destroy(ctx_);
+  }
+
+  unsigned int get_chunk_count() const override { return k_ + m_; }
+  unsigned int get_data_chunk_count() const override { return k_; }
+  size_t get_minimum_granularity() override { return 16; }
+
+  plugin_flags get_supported_optimizations() const override {
+    // the subset the conformance tests verify (cf. ErasureCodeIsa.h:68-79;
+    // CRC composition is not implemented yet, so not claimed)
+    return FLAG_EC_PLUGIN_PARTIAL_READ_OPTIMIZATION |
+           FLAG_EC_PLUGIN_PARTIAL_WRITE_OPTIMIZATION |
+           FLAG_EC_PLUGIN_ZERO_INPUT_ZERO_OUTPUT_OPTIMIZATION |
+           FLAG_EC_PLUGIN_PARITY_DELTA_OPTIMIZATION |
+           FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED |
+           FLAG_EC_PLUGIN_DIRECT_READS;
+  }
+
+  int parse(ErasureCodeProfile &profile, std::ostream *ss) {
+    int err = ErasureCode::parse(profile, ss);
+    err |= to_int("k", profile, &k_, "8", ss);
+    err |= to_int("m", profile, &m_, "3", ss);
+    err |= to_int("w", profile, &w_, "8", ss);
+    err |= to_int("mi355x-device", profile, &device_, "0", ss);
+    err |= to_int("mi355x-streams", profile, &streams_, "2", ss);
+    err |= sanity_check_k_m(k_, m_, ss);
+    if (w_ != 8) {
+      if (ss) *ss << "mi355x: w=" << w_ << " must be 8\n";
+      err = -EINVAL;
+    }
+    if (technique_id(technique_) < 0) {
+      if (ss) *ss << "mi355x: unknown technique " << technique_ << "\n";
+      err = -EINVAL;
+    }
+    profile["technique"] = technique_;
+    return err;
+  }
+
+  int init(ErasureCodeProfile &profile, std::ostream *ss) override {
+    int err = parse(profile, ss);
+    if (err) return err;
+    int r = ecx_create(k_, m_, technique_id(technique_), device_, streams_,
+                       &ctx_);
+    if (r != ECX_OK) {
+      if (ss)
+        *ss << "mi355x: ecx_create failed (" << r
+            << (r == ECX_ERR_NO_GPU ? ": no GPU — this plugin has no CPU "
+                                      "fallback"
+                                    : "")
+            << ")\n";
+      return r;
+    }
+    return ErasureCode::init(profile, ss);
+  }
+
+  unsigned int get_chunk_size(unsigned int stripe_width) const override {
+    return ecx_chunk_size(ctx_, stripe_width);
+  }
+
+  int encode_chunks(const shard_id_map<buffer> &in,
+                    shard_id_map<buffer> &out) override {
+    // marshalling mirror of ErasureCodeIsa.cc:118-165: absent shards are
+    // zeros (NULL pointer convention of the C-ABI); sizes must agree
+    size_t size = 0;
+    const uint8_t *data[64] = {};
+    uint8_t *parity[64] = {};
+    for (auto &&[shard, b] : in) {
+      if ((int)shard >= k_) return -EINVAL;
+      if (!size) size = b.length();
+      else if (size != b.length()) return -EINVAL;
+      data[(int)shard] = b.c_str();
+    }
+    for (auto &&[shard, b] : out) {
+      if ((int)shard < k_) continue;
+      if (!size) size = b.length();
+      else if (size != b.length()) return -EINVAL;
+      parity[(int)shard - k_] = b.c_str();
+    }
+    if (!size) return 0;
+    return ecx_encode_chunks_host(ctx_, data, parity, size);
+  }
+
+  int decode_chunks(const shard_id_set &want_to_read,
+                    shard_id_map<buffer> &in,
+                    shard_id_map<buffer> &out) override {
+    (void)want_to_read;  // all erasures in `out` are reconstructed
+    size_t size = 0;
+    uint8_t *chunks[64] = {};
+    uint64_t present = 0;
+    std::vector<buffer> temps;
+    for (auto &&[shard, b] : in) {
+      size = b.length();
+      chunks[(int)shard] = b.c_str();
+      present |= 1ull << (int)shard;
+    }
+    for (auto &&[shard, b] : out) {
+      size = b.length();
+      chunks[(int)shard] = b.c_str();
+      present &= ~(1ull << (int)shard);
+    }
+    if (!size) return 0;
+    return ecx_decode_chunks_host(ctx_, chunks, present, size);
+  }
+
+  void encode_delta(const buffer &old_data, const buffer &new_data,
+                    buffer *delta_maybe_in_place) override {
+    ecx_encode_delta_host(ctx_, old_data.c_str(), new_data.c_str(),
+                          delta_maybe_in_place->c_str(),
+                          delta_maybe_in_place->length());
+  }
+
+  void apply_delta(const shard_id_map<buffer> &in,
+                   shard_id_map<buffer> &out) override {
+    // loop structure mirrors isa apply_delta (ErasureCodeIsa.cc:333-366)
+    for (auto &&[datashard, databuf] : in) {
+      if ((int)datashard >= k_) continue;
+      for (auto &&[codingshard, codingbuf] : out) {
+        if ((int)codingshard < k_) continue;
+        ecx_apply_delta_host(ctx_, databuf.c_str(), (int)datashard,
+                             (int)codingshard,
+                             const_cast<uint8_t *>(codingbuf.c_str()),
+                             codingbuf.length());
+      }
+    }
+  }
+};
+
+class ErasureCodePluginMi355x final : public ErasureCodePlugin {
+ public:
+  int factory(const std::string &, ErasureCodeProfile &profile,
+              ErasureCodeInterfaceRef *erasure_code,
+              std::ostream *ss) override {
+    std::string technique = "reed_sol_van";
+    if (auto it = profile.find("technique"); it != profile.end())
+      technique = it->second;
+    auto interface = std::make_shared<ErasureCodeMi355x>(technique);
+    int r = interface->init(profile, ss);
+    if (r) return r;
+    *erasure_code = interface;
+    return 0;
+  }
+};
+
+}  // namespace
+
+// plugin C entry points (pattern: ErasureCodePluginJerasure.cc:74-90)
+extern "C" {
+
+const char *__erasure_code_version() { return ECX_HARNESS_VERSION; }
+
+int __erasure_code_init(const char *plugin_name, const char *) {
+  auto &instance = ErasureCodePluginRegistry::instance();
+  auto plugin = std::make_unique<ErasureCodePluginMi355x>();
+  int r = instance.add(plugin_name, plugin.get());
+  if (r == 0) plugin.release();
+  return r;
+}
+}

@@ -1,0 +1,157 @@
+// registry_selftest.cc — exercises the plugin registry's dlopen semantics
+// and the oracle fixture plugin's interface conformance on CPU. Mirrors the
+// reference's registry tests (src/test/erasure-code/TestErasureCodePlugin.cc
+// failure modes) and the round-trip style of TestErasureCodeJerasure.cc.
+// Exit code 0 = all checks passed; prints one line per check.
+#include <cerrno>
+#include <cstring>
+#include <iostream>
+#include <random>
+#include <sstream>
+
+#include "erasure_code_plugin.h"
+
+using namespace ecx;
+
+static int failures = 0;
+
+#define CHECK(cond, name)                                       \
+  do {                                                          \
+    if (cond) {                                                 \
+      std::cout << "ok " << name << "\n";                       \
+    } else {                                                    \
+      std::cout << "FAIL " << name << "\n";                     \
+      failures++;                                               \
+    }                                                           \
+  } while (0)
+
+int main(int argc, char **argv) {
+  std::string dir = argc > 1 ? argv[1] : ".";
+  auto &reg = ErasureCodePluginRegistry::instance();
+  reg.disable_dlclose = true;
+  std::stringstream ss;
+
+  // --- failure modes (ErasureCodePlugin.cc:138-206 semantics) ---
+  {
+    ErasureCodeProfile p{{"k", "2"}, {"m", "1"}};
+    ErasureCodeInterfaceRef ec;
+    int r = reg.factory("does_not_exist", dir, p, &ec, &ss);
+    CHECK(r == -EIO, "dlopen-missing-file => -EIO");
+  }
+  {
+    ErasureCodeProfile p{{"k", "2"}, {"m", "1"}};
+    ErasureCodeInterfaceRef ec;
+    int r = reg.factory("fix_missing_version", dir, p, &ec, &ss);
+    CHECK(r == -EXDEV, "missing __erasure_code_version => -EXDEV");
+    r = reg.factory("fix_bad_version", dir, p, &ec, &ss);
+    CHECK(r == -EXDEV, "wrong version string => -EXDEV");
+    r = reg.factory("fix_missing_init", dir, p, &ec, &ss);
+    CHECK(r == -ENOENT, "missing __erasure_code_init => -ENOENT");
+    r = reg.factory("fix_fail_init", dir, p, &ec, &ss);
+    CHECK(r == -ESRCH, "init failure propagates");
+    r = reg.factory("fix_no_register", dir, p, &ec, &ss);
+    CHECK(r == -EBADF, "init that does not register => -EBADF");
+  }
+
+  // --- oracle fixture plugin: load, profile echo, round trip ---
+  ErasureCodeInterfaceRef ec;
+  {
+    ErasureCodeProfile p{{"k", "4"}, {"m", "2"},
+                         {"technique", "reed_sol_van"}};
+    int r = reg.factory("oracle", dir, p, &ec, &ss);
+    CHECK(r == 0, "oracle plugin factory");
+    if (r != 0) {
+      std::cerr << ss.str() << "\n";
+      return 1;
+    }
+    CHECK(p.at("k") == "4" && p.count("crush-root") == 1,
+          "profile defaults echoed back");
+    CHECK(ec->get_chunk_count() == 6, "chunk count");
+  }
+  {
+    // encode/decode round trip with every 2-erasure pattern
+    const unsigned C = ec->get_chunk_size(4 * 4096);
+    buffer in = buffer::create_aligned(4 * 4096);
+    std::mt19937_64 rng(0xEC);
+    for (size_t i = 0; i < in.length() / 8; i++)
+      ((uint64_t *)in.c_str())[i] = rng();
+    shard_id_set want;
+    for (int i = 0; i < 6; i++) want.insert(i);
+    shard_id_map<buffer> encoded(6);
+    int r = ec->encode(want, in, &encoded);
+    CHECK(r == 0 && encoded.size() == 6 && encoded.at(0).length() == C,
+          "encode");
+    // systematic prefix: data chunks hold the input verbatim
+    CHECK(std::memcmp(encoded.at(0).c_str(), in.c_str(), C) == 0 &&
+              std::memcmp(encoded.at(3).c_str(), in.c_str() + 3 * C,
+                          in.length() - 3 * C) == 0,
+          "systematic prefix");
+    bool all_ok = true;
+    for (int e1 = 0; e1 < 6; e1++)
+      for (int e2 = e1 + 1; e2 < 6; e2++) {
+        shard_id_map<buffer> chunks = encoded;
+        chunks.erase(e1);
+        chunks.erase(e2);
+        shard_id_map<buffer> decoded(6);
+        shard_id_set want_read;
+        want_read.insert(e1);
+        want_read.insert(e2);
+        if (ec->decode(want_read, chunks, &decoded, 0) != 0) all_ok = false;
+        for (int e : {e1, e2})
+          if (std::memcmp(decoded.at(e).c_str(), encoded.at(e).c_str(), C))
+            all_ok = false;
+      }
+    CHECK(all_ok, "exhaustive 2-erasure decode round trip");
+  }
+  {
+    // parity-delta conformance: delta path == full re-encode
+    ErasureCodeProfile p{{"k", "3"}, {"m", "2"},
+                         {"technique", "cauchy"}};
+    ErasureCodeInterfaceRef ec2;
+    int r = reg.factory("oracle", dir, p, &ec2, &ss);
+    CHECK(r == 0, "oracle cauchy factory");
+    unsigned C = 4096;
+    std::mt19937_64 rng(1);
+    shard_id_map<buffer> in(5), out(5);
+    for (int i = 0; i < 3; i++) {
+      buffer b = buffer::create_aligned(C);
+      for (size_t w = 0; w < C / 8; w++) ((uint64_t *)b.c_str())[w] = rng();
+      in[i] = b;
+    }
+    for (int j = 3; j < 5; j++) out[j] = buffer::create_aligned(C);
+    ec2->encode_chunks(in, out);
+    // mutate chunk 1 via delta
+    buffer newc = buffer::create_aligned(C);
+    for (size_t w = 0; w < C / 8; w++) ((uint64_t *)newc.c_str())[w] = rng();
+    buffer delta = buffer::create_aligned(C);
+    ec2->encode_delta(in.at(1), newc, &delta);
+    shard_id_map<buffer> din(5), dout(5);
+    din[1] = delta;
+    dout[3] = out.at(3);
+    dout[4] = out.at(4);
+    ec2->apply_delta(din, dout);
+    in[1] = newc;
+    shard_id_map<buffer> out2(5);
+    for (int j = 3; j < 5; j++) out2[j] = buffer::create_aligned(C);
+    ec2->encode_chunks(in, out2);
+    bool same = !std::memcmp(out.at(3).c_str(), out2.at(3).c_str(), C) &&
+                !std::memcmp(out.at(4).c_str(), out2.at(4).c_str(), C);
+    CHECK(same, "parity delta == re-encode");
+  }
+  {
+    // minimum_to_decode semantics (ErasureCode.cc:154-170)
+    shard_id_set want, avail, minimum;
+    want.insert(0);
+    for (int i : {0, 1, 2, 3, 4, 5}) avail.insert(i);
+    ec->minimum_to_decode(want, avail, minimum, nullptr);
+    CHECK(minimum.size() == 1 && minimum.contains(0),
+          "minimum == want when available");
+    shard_id_set avail2, min2;
+    for (int i : {1, 2, 4, 5}) avail2.insert(i);
+    ec->minimum_to_decode(want, avail2, min2, nullptr);
+    CHECK(min2.size() == 4, "minimum == first k available on erasure");
+  }
+
+  std::cout << (failures ? "FAILURES: " : "all ok: ") << failures << "\n";
+  return failures ? 1 : 0;
+}
