@@ -72,7 +72,12 @@ __device__ __forceinline__ uint32_t ecx_lut(uint32_t t0, uint32_t t1,
   return __builtin_amdgcn_perm(t1, t0, i7) ^ __builtin_amdgcn_perm(0u, t8, i8);
 }
 
-template <int NOUT, bool ACCUM>
+// VPT = 16-byte vectors per thread per grid-stride iteration. VPT=2 gives
+// each lane two independent load/accumulate chains (more memory-level
+// parallelism, half the loop overhead); both vectors of a lane are
+// blockDim.x apart so every wave access stays a fully coalesced 1 KiB
+// transaction.
+template <int NOUT, bool ACCUM, int VPT>
 __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const EcLaunchParams* __restrict__ pb, long chunk_bytes,
@@ -96,32 +101,47 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
   const uint8_t* sbase = buf + stripe * chunks_per_stripe * chunk_bytes;
   uint8_t* obase = obuf + stripe * chunks_per_stripe * chunk_bytes;
 
-  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x;
-       p < vecs_per_chunk; p += (long)gridDim.x * blockDim.x) {
-    const long off = p << 4;
-    uint32_t acc[NOUT][4];
+  for (long p0 = (long)blockIdx.x * blockDim.x * VPT + threadIdx.x;
+       p0 < vecs_per_chunk; p0 += (long)gridDim.x * blockDim.x * VPT) {
+    long off[VPT];
+    bool live[VPT];
 #pragma unroll
-    for (int j = 0; j < NOUT; j++) {
-      if (ACCUM) {
-        const uint4 o =
-            *reinterpret_cast<const uint4*>(obase + (long)s_out[j] * chunk_bytes + off);
-        acc[j][0] = o.x; acc[j][1] = o.y; acc[j][2] = o.z; acc[j][3] = o.w;
-      } else {
-        acc[j][0] = acc[j][1] = acc[j][2] = acc[j][3] = 0u;
-      }
+    for (int v = 0; v < VPT; v++) {
+      long p = p0 + (long)v * blockDim.x;
+      live[v] = p < vecs_per_chunk;
+      off[v] = p << 4;
     }
+    uint32_t acc[NOUT][VPT][4];
+#pragma unroll
+    for (int j = 0; j < NOUT; j++)
+#pragma unroll
+      for (int v = 0; v < VPT; v++) {
+        if (ACCUM && live[v]) {
+          const uint4 o = *reinterpret_cast<const uint4*>(
+              obase + (long)s_out[j] * chunk_bytes + off[v]);
+          acc[j][v][0] = o.x; acc[j][v][1] = o.y;
+          acc[j][v][2] = o.z; acc[j][v][3] = o.w;
+        } else {
+          acc[j][v][0] = acc[j][v][1] = acc[j][v][2] = acc[j][v][3] = 0u;
+        }
+      }
 
     for (int i = 0; i < n_src; i++) {
-      const uint4 d =
-          *reinterpret_cast<const uint4*>(sbase + (long)s_src[i] * chunk_bytes + off);
-      const uint32_t dq[4] = {d.x, d.y, d.z, d.w};
-      uint32_t i7l[4], i8l[4], i7h[4], i8h[4];
+      const uint8_t* sp = sbase + (long)s_src[i] * chunk_bytes;
+      uint32_t dq[VPT][4];
+      uint32_t i7l[VPT][4], i8l[VPT][4], i7h[VPT][4], i8h[VPT][4];
 #pragma unroll
-      for (int q = 0; q < 4; q++) {
-        i7l[q] = dq[q] & 0x07070707u;
-        i8l[q] = (dq[q] >> 3) & 0x01010101u;
-        i7h[q] = (dq[q] >> 4) & 0x07070707u;
-        i8h[q] = (dq[q] >> 7) & 0x01010101u;
+      for (int v = 0; v < VPT; v++) {
+        uint4 d = {0, 0, 0, 0};
+        if (live[v]) d = *reinterpret_cast<const uint4*>(sp + off[v]);
+        dq[v][0] = d.x; dq[v][1] = d.y; dq[v][2] = d.z; dq[v][3] = d.w;
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+          i7l[v][q] = dq[v][q] & 0x07070707u;
+          i8l[v][q] = (dq[v][q] >> 3) & 0x01010101u;
+          i7h[v][q] = (dq[v][q] >> 4) & 0x07070707u;
+          i8h[v][q] = (dq[v][q] >> 7) & 0x01010101u;
+        }
       }
 #pragma unroll
       for (int j = 0; j < NOUT; j++) {
@@ -131,25 +151,34 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
         if (cls == 0) continue;
         if (cls == 1) {
 #pragma unroll
-          for (int q = 0; q < 4; q++) acc[j][q] ^= dq[q];
+          for (int v = 0; v < VPT; v++)
+#pragma unroll
+            for (int q = 0; q < 4; q++) acc[j][v][q] ^= dq[v][q];
         } else {
           const uint32_t* T = &s_tabs[(j * n_src + i) * 6];
           const uint32_t t0 = T[0], t1 = T[1], t2 = T[2], t3 = T[3],
                          t4 = T[4], t5 = T[5];
 #pragma unroll
-          for (int q = 0; q < 4; q++)
-            acc[j][q] ^= ecx_lut(t0, t1, t2, i7l[q], i8l[q]) ^
-                         ecx_lut(t3, t4, t5, i7h[q], i8h[q]);
+          for (int v = 0; v < VPT; v++)
+#pragma unroll
+            for (int q = 0; q < 4; q++)
+              acc[j][v][q] ^= ecx_lut(t0, t1, t2, i7l[v][q], i8l[v][q]) ^
+                              ecx_lut(t3, t4, t5, i7h[v][q], i8h[v][q]);
         }
       }
     }
 
 #pragma unroll
-    for (int j = 0; j < NOUT; j++) {
-      uint4 o;
-      o.x = acc[j][0]; o.y = acc[j][1]; o.z = acc[j][2]; o.w = acc[j][3];
-      *reinterpret_cast<uint4*>(obase + (long)s_out[j] * chunk_bytes + off) = o;
-    }
+    for (int j = 0; j < NOUT; j++)
+#pragma unroll
+      for (int v = 0; v < VPT; v++) {
+        if (!live[v]) continue;
+        uint4 o;
+        o.x = acc[j][v][0]; o.y = acc[j][v][1];
+        o.z = acc[j][v][2]; o.w = acc[j][v][3];
+        *reinterpret_cast<uint4*>(obase + (long)s_out[j] * chunk_bytes +
+                                  off[v]) = o;
+      }
   }
 }
 
@@ -450,26 +479,43 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
   HIP_TRY(hipEventRecord(s.ev_param, s.stream));
 
   const long vecs = (long)(chunk_bytes >> 4);
-  // >= 8 vectors per thread per block pass; cap blocks-x; floor so small
-  // chunks still fill the chip via the y (stripe) dimension.
-  int gx = (int)std::min<long>((vecs + 256 * 8 - 1) / (256 * 8), 1024);
+  // Tunables (A/B-able via env on the GPU box): VPT = 16B vectors per
+  // thread per iteration, TILE = target vectors per thread per launch.
+  static const int env_vpt = [] {
+    const char* v = getenv("ECX_VPT");
+    int x = v ? atoi(v) : 2;
+    return (x == 1 || x == 2) ? x : 2;
+  }();
+  static const int env_tile = [] {
+    const char* v = getenv("ECX_TILE");
+    int x = v ? atoi(v) : 8;
+    return x >= 1 ? x : 8;
+  }();
+  const int vpt = (vecs >= 2 * 256) ? env_vpt : 1;
+  const long per_block = 256L * vpt;
+  long tiles = (vecs + per_block - 1) / per_block;
+  long loops = std::max(1, env_tile / vpt);
+  int gx = (int)std::min<long>((tiles + loops - 1) / loops, 1024);
   if (gx < 1) gx = 1;
   // if few stripes, widen x so total blocks cover 256 CUs * a few waves
-  while ((long)gx * n_stripes < 2048 && gx < (vecs + 255) / 256) gx *= 2;
+  while ((long)gx * n_stripes < 2048 && gx < tiles) gx *= 2;
   dim3 grid(gx, (unsigned)n_stripes);
 
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
   const int cps = ctx->k + ctx->m;
-#define ECX_DISPATCH(NO)                                                      \
-  case NO:                                                                    \
-    if (accum)                                                                \
-      hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, true>), grid, dim3(256), 0, \
-                         s.stream, d_buf, d_obuf, s.d_params,                 \
-                         (long)chunk_bytes, cps, vecs);                       \
-    else                                                                      \
-      hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, false>), grid, dim3(256),   \
-                         0, s.stream, d_buf, d_obuf, s.d_params,              \
-                         (long)chunk_bytes, cps, vecs);                       \
+#define ECX_LAUNCH(NO, AC, VP)                                               \
+  hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP>), grid, dim3(256), 0,  \
+                     s.stream, d_buf, d_obuf, s.d_params, (long)chunk_bytes, \
+                     cps, vecs)
+#define ECX_DISPATCH(NO)                      \
+  case NO:                                    \
+    if (accum) {                              \
+      if (vpt == 2) ECX_LAUNCH(NO, true, 2);  \
+      else ECX_LAUNCH(NO, true, 1);           \
+    } else {                                  \
+      if (vpt == 2) ECX_LAUNCH(NO, false, 2); \
+      else ECX_LAUNCH(NO, false, 1);          \
+    }                                         \
     break;
   switch (params.n_out) {
     ECX_DISPATCH(1)
@@ -480,6 +526,7 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
       return ECX_ERR_INVAL;
   }
 #undef ECX_DISPATCH
+#undef ECX_LAUNCH
   HIP_TRY(hipGetLastError());
   if (time_it) {
     HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
