@@ -481,15 +481,18 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
   const long vecs = (long)(chunk_bytes >> 4);
   // Tunables (A/B-able via env on the GPU box): VPT = 16B vectors per
   // thread per iteration, TILE = target vectors per thread per launch.
+  // Defaults from the round-1 sweep (profiles/rocprof_r01_summary.md):
+  // VPT=1 keeps 8 waves/SIMD at NOUT=3 and beat VPT=2; TILE=4 won the
+  // grid-shape sweep.
   static const int env_vpt = [] {
     const char* v = getenv("ECX_VPT");
-    int x = v ? atoi(v) : 2;
-    return (x == 1 || x == 2) ? x : 2;
+    int x = v ? atoi(v) : 1;
+    return (x == 1 || x == 2) ? x : 1;
   }();
   static const int env_tile = [] {
     const char* v = getenv("ECX_TILE");
-    int x = v ? atoi(v) : 8;
-    return x >= 1 ? x : 8;
+    int x = v ? atoi(v) : 4;
+    return x >= 1 ? x : 4;
   }();
   const int vpt = (vecs >= 2 * 256) ? env_vpt : 1;
   const long per_block = 256L * vpt;
