@@ -112,36 +112,34 @@ static int have_avx2(void)
 static int have_avx2(void) { return 0; }
 #endif
 
-/* One stripe: parity[j] = sum_i M[j][i]*data[i], blocked for cache reuse. */
-static void encode_stripe(int k, int m, const uint8_t *tables /* m*k*32 */,
-                          const uint8_t *coding_rows,
-                          const uint8_t *const *data, uint8_t *const *parity,
-                          size_t len)
+#define ECCPU_BLK 16384
+
+/* One cache-blocked region: parity[j][off..off+n) for all j. */
+static void encode_block(int k, int m, const uint8_t *tables /* m*k*32 */,
+                         const uint8_t *coding_rows,
+                         const uint8_t *const *data, uint8_t *const *parity,
+                         size_t off, size_t n)
 {
-  const size_t BLK = 16384;
-  for (size_t off = 0; off < len; off += BLK) {
-    size_t n = len - off < BLK ? len - off : BLK;
-    for (int j = 0; j < m; j++) {
-      uint8_t *out = parity[j] + off;
-      memset(out, 0, n);
-      for (int i = 0; i < k; i++) {
-        if (data[i] == NULL)
-          continue;
-        uint8_t c = coding_rows[(size_t)j * k + i];
-        if (c == 0)
-          continue;
+  for (int j = 0; j < m; j++) {
+    uint8_t *out = parity[j] + off;
+    memset(out, 0, n);
+    for (int i = 0; i < k; i++) {
+      if (data[i] == NULL)
+        continue;
+      uint8_t c = coding_rows[(size_t)j * k + i];
+      if (c == 0)
+        continue;
 #if defined(__x86_64__)
-        if (have_avx2()) {
-          if (c == 1)
-            region_xor_acc_avx2(data[i] + off, out, n);
-          else
-            region_mul_acc_avx2(tables + ((size_t)j * k + i) * 32,
-                                data[i] + off, out, n);
-          continue;
-        }
-#endif
-        region_mul_acc_scalar(c, data[i] + off, out, n);
+      if (have_avx2()) {
+        if (c == 1)
+          region_xor_acc_avx2(data[i] + off, out, n);
+        else
+          region_mul_acc_avx2(tables + ((size_t)j * k + i) * 32,
+                              data[i] + off, out, n);
+        continue;
       }
+#endif
+      region_mul_acc_scalar(c, data[i] + off, out, n);
     }
   }
 }
@@ -170,10 +168,15 @@ int eccpu_encode_batch(int technique, int k, int m, uint8_t *base,
     for (int i = 0; i < k; i++)
       mul_table_32(rows[(size_t)j * k + i], tables + ((size_t)j * k + i) * 32);
 
+  /* flatten (stripe, block) so many threads stay fed even for small
+   * stripe counts (the GPU-box host has 256 hardware threads) */
+  long bpc = (long)((chunk_bytes + ECCPU_BLK - 1) / ECCPU_BLK);
+  long total = n_stripes * bpc;
 #ifdef _OPENMP
-#pragma omp parallel for schedule(dynamic, 1)
+#pragma omp parallel for schedule(dynamic, 4)
 #endif
-  for (long s = 0; s < n_stripes; s++) {
+  for (long w = 0; w < total; w++) {
+    long s = w / bpc, b = w % bpc;
     const uint8_t *data[255];
     uint8_t *parity[255];
     uint8_t *stripe = base + (size_t)s * (k + m) * chunk_bytes;
@@ -181,7 +184,9 @@ int eccpu_encode_batch(int technique, int k, int m, uint8_t *base,
       data[i] = stripe + (size_t)i * chunk_bytes;
     for (int j = 0; j < m; j++)
       parity[j] = stripe + (size_t)(k + j) * chunk_bytes;
-    encode_stripe(k, m, tables, rows, data, parity, chunk_bytes);
+    size_t off = (size_t)b * ECCPU_BLK;
+    size_t n = chunk_bytes - off < ECCPU_BLK ? chunk_bytes - off : ECCPU_BLK;
+    encode_block(k, m, tables, rows, data, parity, off, n);
   }
   return 0;
 }
@@ -238,10 +243,13 @@ int eccpu_decode_batch(int technique, int k, int m, uint8_t *base,
     for (int i = 0; i < k; i++)
       mul_table_32(c[(size_t)j * k + i], tables + ((size_t)j * k + i) * 32);
 
+  long bpc = (long)((chunk_bytes + ECCPU_BLK - 1) / ECCPU_BLK);
+  long total = n_stripes * bpc;
 #ifdef _OPENMP
-#pragma omp parallel for schedule(dynamic, 1)
+#pragma omp parallel for schedule(dynamic, 4)
 #endif
-  for (long s = 0; s < n_stripes; s++) {
+  for (long w = 0; w < total; w++) {
+    long s = w / bpc, b = w % bpc;
     const uint8_t *src[255];
     uint8_t *dst[255];
     uint8_t *stripe = base + (size_t)s * (k + m) * chunk_bytes;
@@ -249,7 +257,9 @@ int eccpu_decode_batch(int technique, int k, int m, uint8_t *base,
       src[i] = stripe + (size_t)decode_index[i] * chunk_bytes;
     for (int p = 0; p < nerrs; p++)
       dst[p] = stripe + (size_t)erasures[p] * chunk_bytes;
-    encode_stripe(k, nerrs, tables, c, src, dst, chunk_bytes);
+    size_t off = (size_t)b * ECCPU_BLK;
+    size_t n = chunk_bytes - off < ECCPU_BLK ? chunk_bytes - off : ECCPU_BLK;
+    encode_block(k, nerrs, tables, c, src, dst, off, n);
   }
   return 0;
 }
