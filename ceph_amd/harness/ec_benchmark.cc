@@ -93,12 +93,14 @@ int Bench::setup(int argc, char **argv) {
       default: return -EINVAL;
     }
   }
+  // k/m usually come from the profile (benchmark.cc:135-141); for plugins
+  // whose profile omits them (LRC with explicit layers) they are derived
+  // from the constructed codec after factory().
   try {
     k = std::stoi(profile.at("k"));
     m = std::stoi(profile.at("m"));
   } catch (const std::exception &) {
-    std::cerr << "profile must set k and m (-P k=.. -P m=..)\n";
-    return -EINVAL;
+    k = m = 0;
   }
   return 0;
 }
@@ -129,6 +131,10 @@ int Bench::encode() {
   if (code) {
     std::cerr << messages.str() << std::endl;
     return code;
+  }
+  if (k <= 0) {
+    k = erasure_code->get_data_chunk_count();
+    m = erasure_code->get_chunk_count() - k;
   }
   buffer in = make_input();
   shard_id_set want_to_encode;
@@ -190,6 +196,10 @@ int Bench::decode() {
   if (code) {
     std::cerr << messages.str() << std::endl;
     return code;
+  }
+  if (k <= 0) {
+    k = erasure_code->get_data_chunk_count();
+    m = erasure_code->get_chunk_count() - k;
   }
   buffer in = make_input();
   shard_id_set want_all;

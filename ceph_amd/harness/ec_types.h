@@ -37,6 +37,7 @@ class shard_id_set {
   void insert(shard_id_t s) { bits_[(uint8_t)s.id >> 6] |= 1ull << (s.id & 63); }
   void insert(int s) { insert(shard_id_t((int8_t)s)); }
   void erase(shard_id_t s) { bits_[(uint8_t)s.id >> 6] &= ~(1ull << (s.id & 63)); }
+  void erase(int s) { erase(shard_id_t((int8_t)s)); }
   bool contains(shard_id_t s) const {
     return bits_[(uint8_t)s.id >> 6] >> (s.id & 63) & 1;
   }

@@ -107,10 +107,17 @@ class ErasureCodeMi355x final : public ErasureCode {
       data[(int)shard] = b.c_str();
     }
     for (auto &&[shard, b] : out) {
-      if ((int)shard < k_) continue;
       if (!size) size = b.length();
       else if (size != b.length()) return -EINVAL;
-      parity[(int)shard - k_] = b.c_str();
+      if ((int)shard < k_) {
+        // a data-position buffer arriving via the out map is a SOURCE —
+        // the LRC local-layer pattern feeds upper-layer parity back in as
+        // data this way (ErasureCodeLrc.cc:985-991 + isa marshalling
+        // ErasureCodeIsa.cc:134-141 treat in/out symmetrically)
+        data[(int)shard] = b.c_str();
+      } else {
+        parity[(int)shard - k_] = b.c_str();
+      }
     }
     if (!size) return 0;
     return ecx_encode_chunks_host(ctx_, data, parity, size);
@@ -135,6 +142,15 @@ class ErasureCodeMi355x final : public ErasureCode {
       present &= ~(1ull << (int)shard);
     }
     if (!size) return 0;
+    // chunks in neither map are erasures reconstructed into scratch and
+    // discarded (the reference invents such buffers,
+    // ErasureCodeJerasure.cc:230-240)
+    for (int i = 0; i < k_ + m_; i++) {
+      if (!chunks[i] && !(present & (1ull << i))) {
+        temps.push_back(buffer::create_aligned(size));
+        chunks[i] = temps.back().c_str();
+      }
+    }
     return ecx_decode_chunks_host(ctx_, chunks, present, size);
   }
 

@@ -80,8 +80,12 @@ class ErasureCodeOracle final : public ErasureCode {
     std::vector<buffer> scratch;
     for (int j = 0; j < m_; j++) parity[j] = nullptr;
     for (auto &&[shard, b] : out) {
-      if ((int)shard < k_) continue;
       if (!size) size = b.length();
+      if ((int)shard < k_) {
+        // data position via the out map == source (LRC local-layer case)
+        data[(int)shard] = b.c_str();
+        continue;
+      }
       parity[(int)shard - k_] = b.c_str();
     }
     // parity chunks not requested still need computing space
