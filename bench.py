@@ -108,6 +108,132 @@ def cpu_baseline(args, budget_s=12.0):
     }
 
 
+def bench_mixed(args):
+    """BASELINE configs[4]: mixed k/m and 64 KiB..4 MiB chunks, streamed
+    encode+decode with a per-(sub-batch) latency histogram. Shards across
+    ranks like the main bench (weak scaling: every rank runs the full shape
+    set on its own GPU)."""
+    import ceph_amd
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        torch.cuda.set_device(local_rank)
+        tdist.init_process_group("nccl")
+        dist = tdist
+
+    # (k, m, chunk_bytes): BASELINE's k in {4,6,8,12}, m in {2,3,4},
+    # chunks 64 KiB..4 MiB; sub-batch sized ~2 GiB of data each
+    shapes = [(4, 2, 64 << 10), (6, 3, 256 << 10), (8, 3, 1 << 20),
+              (12, 4, 4 << 20)]
+    rng = np.random.default_rng(args.seed)
+    ctxs, bufs = [], []
+    for (k, m, C) in shapes:
+        S = max(8, int(2 * GIB // (k * C)))
+        ctx = ceph_amd.EcContext(k, m, "reed_sol_van", device=local_rank,
+                                 n_streams=args.streams)
+        nbytes = S * (k + m) * C
+        d = ctx.dbuf_alloc(nbytes)
+        ctx.fill_random(d, nbytes, args.seed + rank)
+        ctx.sync()
+        n = k + m
+        er = sorted(rng.choice(n, size=min(m, 3), replace=False).tolist())
+        mask = (1 << n) - 1
+        for e in er:
+            mask &= ~(1 << e)
+        ctxs.append((ctx, d, k, m, C, S, mask))
+
+    def step(lat=None):
+        for (ctx, d, k, m, C, S, mask) in ctxs:
+            t0 = time.perf_counter()
+            ctx.encode_batch(d, S, C)
+            ctx.decode_batch(d, S, C, mask)
+            ctx.sync()
+            if lat is not None:
+                lat.append((k, m, C, (time.perf_counter() - t0) * 1e3))
+
+    for _ in range(args.warmup):
+        step()
+    if dist:
+        import torch
+        dist.barrier()
+        torch.cuda.synchronize()
+    lat = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step(lat)
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+        torch.cuda.synchronize()
+        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        bytes_per_step = sum(2 * k * C * S for (_, _, k, m, C, S, _) in ctxs)
+        ms = sorted(x[3] for x in lat)
+        hist = {
+            "p10_ms": round(ms[int(len(ms) * .10)], 3),
+            "p50_ms": round(ms[len(ms) // 2], 3),
+            "p90_ms": round(ms[int(len(ms) * .90)], 3),
+            "p99_ms": round(ms[min(len(ms) - 1, int(len(ms) * .99))], 3),
+            "per_shape_ms": {f"k{k}m{m}c{C//1024}k": round(
+                float(np.mean([x[3] for x in lat if x[:3] == (k, m, C)])), 3)
+                for (k, m, C) in {x[:3] for x in lat}},
+        }
+        value = world * bytes_per_step * args.steps / GIB / elapsed
+        print(json.dumps({
+            "metric": "EC encode+decode GiB/s",
+            "value": round(value, 2), "unit": "GiB/s", "n_gpus": world,
+            "steps": args.steps, "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "u8", "data": "synthetic",
+            "config": {"workload": "mixed k/m 64KiB-4MiB streamed "
+                                   "encode+decode (BASELINE configs[4])",
+                       "shapes": [f"k{k}m{m}c{C >> 10}KiB_s{S}"
+                                  for (_, _, k, m, C, S, _) in ctxs],
+                       "seed": hex(args.seed)},
+            "latency_histogram": hist,
+        }))
+    for (ctx, d, *_rest) in ctxs:
+        ctx.dbuf_free(d)
+        ctx.close()
+    if dist:
+        dist.destroy_process_group()
+
+
+def bench_hostpath(args):
+    """PCIe-inclusive plugin-path probe (single-stripe host-pointer calls,
+    the drop-in path): reported separately from the device-resident metric
+    per DESIGN.md §5 — never the headline value."""
+    import ceph_amd
+    k, m, C = args.k, args.m, args.chunk_bytes
+    ctx = ceph_amd.EcContext(k, m, args.technique, device=0)
+    rng = np.random.default_rng(args.seed)
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    ctx.encode_chunks(data)  # warm
+    t0 = time.perf_counter()
+    iters = 0
+    while time.perf_counter() - t0 < 8.0:
+        ctx.encode_chunks(data)
+        iters += 1
+    dt = time.perf_counter() - t0
+    print(json.dumps({
+        "metric": "EC host-path (PCIe-inclusive) encode GiB/s",
+        "value": round(iters * k * C / GIB / dt, 3), "unit": "GiB/s",
+        "n_gpus": 1, "note": ("single-stripe ecx_encode_chunks_host incl. "
+                              "H2D+D2H staging; drop-in plugin path, not "
+                              "the device-resident metric"),
+        "config": {"k": k, "m": m, "chunk_bytes": C, "iters": iters},
+    }))
+    ctx.close()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -124,14 +250,29 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--no-selfcheck", action="store_true")
     ap.add_argument("--streams", type=int, default=2)
+    ap.add_argument("--config", choices=["rs83", "cauchy104", "mixed",
+                                         "hostpath"],
+                    default="rs83",
+                    help="BASELINE preset: rs83=configs[1] (default), "
+                         "cauchy104=configs[2], mixed=configs[4] shape "
+                         "sweep w/ latency histogram, hostpath=PCIe "
+                         "plugin-path probe")
     args = ap.parse_args()
     args.chunk_bytes = args.chunk_kib * 1024
+    if args.config == "cauchy104":
+        args.k, args.m, args.technique, args.erasures = 10, 4, "cauchy", 4
+        args.stripes = min(args.stripes, 3072)  # 14 chunks/stripe, ~42 GiB
 
     import ceph_amd
 
     if ceph_amd.device_count() < 1:
         print(json.dumps({"error": "no GPU visible; bench requires MI355X"}))
         sys.exit(1)
+
+    if args.config == "mixed":
+        return bench_mixed(args)
+    if args.config == "hostpath":
+        return bench_hostpath(args)
 
     # distributed setup (torchrun provides RANK/WORLD_SIZE/LOCAL_RANK)
     rank = int(os.environ.get("RANK", "0"))
@@ -216,9 +357,12 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": ("RS-Vandermonde k=8 m=3, 1 MiB chunks, "
-                             "4096-stripe batch encode+decode, "
-                             "device-resident (BASELINE configs[1]+[2])"),
+                "workload": (
+                    f"{args.technique} k={k} m={m}, {C >> 10} KiB chunks, "
+                    f"{S}-stripe batch encode+decode({len(erased)} erasures)"
+                    ", device-resident "
+                    + ("(BASELINE configs[2])" if args.config == "cauchy104"
+                       else "(BASELINE configs[1]+[2])")),
                 "k": k, "m": m, "chunk_bytes": C, "stripes_per_gpu": S,
                 "technique": args.technique, "erasures": erased,
                 "seed": hex(args.seed),

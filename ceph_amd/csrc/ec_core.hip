@@ -62,6 +62,8 @@ struct EcLaunchParams {
   uint32_t tabs[ECX_MAX_OUT * ECX_MAX_K * 6];
 };
 
+typedef uint32_t v4u __attribute__((ext_vector_type(4)));
+
 // 16-entry byte lookup over 4 packed bytes:
 //   T7 entries t0 (0..3) / t1 (4..7) selected by i7 in 0..7,
 //   2-entry T8 (byte1 of t8) selected by i8 in 0..1.
@@ -77,7 +79,7 @@ __device__ __forceinline__ uint32_t ecx_lut(uint32_t t0, uint32_t t1,
 // parallelism, half the loop overhead); both vectors of a lane are
 // blockDim.x apart so every wave access stays a fully coalesced 1 KiB
 // transaction.
-template <int NOUT, bool ACCUM, int VPT>
+template <int NOUT, bool ACCUM, int VPT, bool NT>
 __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const EcLaunchParams* __restrict__ pb, long chunk_bytes,
@@ -117,7 +119,7 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
 #pragma unroll
       for (int v = 0; v < VPT; v++) {
         if (ACCUM && live[v]) {
-          const uint4 o = *reinterpret_cast<const uint4*>(
+          const v4u o = *reinterpret_cast<const v4u*>(
               obase + (long)s_out[j] * chunk_bytes + off[v]);
           acc[j][v][0] = o.x; acc[j][v][1] = o.y;
           acc[j][v][2] = o.z; acc[j][v][3] = o.w;
@@ -132,8 +134,11 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
       uint32_t i7l[VPT][4], i8l[VPT][4], i7h[VPT][4], i8h[VPT][4];
 #pragma unroll
       for (int v = 0; v < VPT; v++) {
-        uint4 d = {0, 0, 0, 0};
-        if (live[v]) d = *reinterpret_cast<const uint4*>(sp + off[v]);
+        v4u d = {0, 0, 0, 0};
+        if (live[v]) {
+          const v4u* p4 = reinterpret_cast<const v4u*>(sp + off[v]);
+          d = NT ? __builtin_nontemporal_load(p4) : *p4;
+        }
         dq[v][0] = d.x; dq[v][1] = d.y; dq[v][2] = d.z; dq[v][3] = d.w;
 #pragma unroll
         for (int q = 0; q < 4; q++) {
@@ -173,11 +178,15 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
 #pragma unroll
       for (int v = 0; v < VPT; v++) {
         if (!live[v]) continue;
-        uint4 o;
+        v4u o;
         o.x = acc[j][v][0]; o.y = acc[j][v][1];
         o.z = acc[j][v][2]; o.w = acc[j][v][3];
-        *reinterpret_cast<uint4*>(obase + (long)s_out[j] * chunk_bytes +
-                                  off[v]) = o;
+        v4u* p4 = reinterpret_cast<v4u*>(obase + (long)s_out[j] * chunk_bytes +
+                                         off[v]);
+        if (NT)
+          __builtin_nontemporal_store(o, p4);
+        else
+          *p4 = o;
       }
   }
 }
@@ -494,6 +503,10 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
     int x = v ? atoi(v) : 4;
     return x >= 1 ? x : 4;
   }();
+  static const int env_nt = [] {
+    const char* v = getenv("ECX_NT");
+    return v ? atoi(v) : 0;
+  }();
   const int vpt = (vecs >= 2 * 256) ? env_vpt : 1;
   const long per_block = 256L * vpt;
   long tiles = (vecs + per_block - 1) / per_block;
@@ -506,19 +519,24 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
 
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
   const int cps = ctx->k + ctx->m;
-#define ECX_LAUNCH(NO, AC, VP)                                               \
-  hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP>), grid, dim3(256), 0,  \
-                     s.stream, d_buf, d_obuf, s.d_params, (long)chunk_bytes, \
-                     cps, vecs)
-#define ECX_DISPATCH(NO)                      \
-  case NO:                                    \
-    if (accum) {                              \
-      if (vpt == 2) ECX_LAUNCH(NO, true, 2);  \
-      else ECX_LAUNCH(NO, true, 1);           \
-    } else {                                  \
-      if (vpt == 2) ECX_LAUNCH(NO, false, 2); \
-      else ECX_LAUNCH(NO, false, 1);          \
-    }                                         \
+#define ECX_LAUNCH(NO, AC, VP, NTF)                                          \
+  hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF>), grid,           \
+                     dim3(256), 0, s.stream, d_buf, d_obuf, s.d_params,      \
+                     (long)chunk_bytes, cps, vecs)
+#define ECX_VARIANT(NO, AC)                          \
+  do {                                               \
+    if (vpt == 2) {                                  \
+      if (env_nt) ECX_LAUNCH(NO, AC, 2, true);       \
+      else ECX_LAUNCH(NO, AC, 2, false);             \
+    } else {                                         \
+      if (env_nt) ECX_LAUNCH(NO, AC, 1, true);       \
+      else ECX_LAUNCH(NO, AC, 1, false);             \
+    }                                                \
+  } while (0)
+#define ECX_DISPATCH(NO)              \
+  case NO:                            \
+    if (accum) ECX_VARIANT(NO, true); \
+    else ECX_VARIANT(NO, false);      \
     break;
   switch (params.n_out) {
     ECX_DISPATCH(1)
@@ -529,6 +547,7 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
       return ECX_ERR_INVAL;
   }
 #undef ECX_DISPATCH
+#undef ECX_VARIANT
 #undef ECX_LAUNCH
   HIP_TRY(hipGetLastError());
   if (time_it) {
