@@ -71,6 +71,15 @@ def _lib():
                                         ctypes.c_int, ctypes.c_int,
                                         ctypes.c_void_p, ctypes.c_size_t,
                                         ctypes.c_int]
+    lib.ecx_encode_slices.argtypes = [ctypes.c_void_p,
+                                      ctypes.POINTER(ctypes.c_void_p),
+                                      ctypes.POINTER(ctypes.c_size_t),
+                                      ctypes.c_int, ctypes.c_int]
+    lib.ecx_decode_slices.argtypes = [ctypes.c_void_p,
+                                      ctypes.POINTER(ctypes.c_void_p),
+                                      ctypes.POINTER(ctypes.c_size_t),
+                                      ctypes.c_int, ctypes.c_uint64,
+                                      ctypes.c_int]
     lib.ecx_sync.argtypes = [ctypes.c_void_p, ctypes.c_int]
     lib.ecx_last_kernel_ms.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                        ctypes.POINTER(ctypes.c_double)]
@@ -188,6 +197,28 @@ class EcContext:
     def decode_batch(self, dptr, n_stripes, chunk_bytes, present_mask, slot=0):
         _ck(lib().ecx_decode_batch(self._h, dptr, n_stripes, chunk_bytes,
                                    present_mask, slot), "ecx_decode_batch")
+
+    def _slice_args(self, chunk_ptrs, sizes):
+        n = len(sizes)
+        assert len(chunk_ptrs) == n * (self.k + self.m)
+        arr = (ctypes.c_void_p * len(chunk_ptrs))()
+        for i, p in enumerate(chunk_ptrs):
+            arr[i] = None if p in (None, 0) else int(p)
+        sz = (ctypes.c_size_t * n)(*sizes)
+        return arr, sz, n
+
+    def encode_slices(self, chunk_ptrs, sizes, slot=0):
+        """Variable-size slice batch: chunk_ptrs is a flat list
+        of n*(k+m) device addresses (None => zeros data chunk); sizes is
+        the per-slice byte length (multiples of 16). One kernel launch."""
+        arr, sz, n = self._slice_args(chunk_ptrs, sizes)
+        _ck(lib().ecx_encode_slices(self._h, arr, sz, n, slot),
+            "ecx_encode_slices")
+
+    def decode_slices(self, chunk_ptrs, sizes, present_mask, slot=0):
+        arr, sz, n = self._slice_args(chunk_ptrs, sizes)
+        _ck(lib().ecx_decode_slices(self._h, arr, sz, n, present_mask,
+                                    slot), "ecx_decode_slices")
 
     def sync(self, slot=0):
         _ck(lib().ecx_sync(self._h, slot), "ecx_sync")

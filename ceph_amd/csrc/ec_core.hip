@@ -191,6 +191,92 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
   }
 }
 
+// Variable-size slice batch: one launch over N independent
+// slices of differing length, each with its own k+m device chunk pointers.
+// Blocks are pre-assigned (slice id, vec offset) by the host so there is no
+// per-thread search; a NULL data pointer means a zeros chunk for that slice
+// only (wave-uniform skip).
+template <int NOUT, bool NT>
+__global__ __launch_bounds__(256, 2) void ec_gf_slices_kernel(
+    const uint64_t* __restrict__ chunk_ptrs,  // [n_slices * cps]
+    const int* __restrict__ block_slice,      // [gridDim.x]
+    const long* __restrict__ block_voff,      // [gridDim.x]
+    const long* __restrict__ slice_vecs,      // [n_slices]
+    const EcLaunchParams* __restrict__ pb, int cps, int vecs_per_block) {
+  __shared__ uint32_t s_tabs[ECX_MAX_OUT * ECX_MAX_K * 6];
+  __shared__ int s_src[ECX_MAX_K];
+  __shared__ int s_out[ECX_MAX_OUT];
+  __shared__ uint8_t s_cls[ECX_MAX_OUT * ECX_MAX_K];
+  const int n_src = pb->n_src;
+  for (int t = threadIdx.x; t < NOUT * n_src * 6; t += blockDim.x)
+    s_tabs[t] = pb->tabs[t];
+  for (int t = threadIdx.x; t < n_src; t += blockDim.x)
+    s_src[t] = pb->src_ids[t];
+  for (int t = threadIdx.x; t < NOUT; t += blockDim.x)
+    s_out[t] = pb->out_ids[t];
+  for (int t = threadIdx.x; t < NOUT * n_src; t += blockDim.x)
+    s_cls[t] = pb->cls[t];
+  __syncthreads();
+
+  const int sl = block_slice[blockIdx.x];
+  const long v0 = block_voff[blockIdx.x];
+  const long nv = slice_vecs[sl];
+  const long vend = v0 + vecs_per_block < nv ? v0 + vecs_per_block : nv;
+  const uint64_t* ptrs = chunk_ptrs + (long)sl * cps;
+
+  for (long p = v0 + threadIdx.x; p < vend; p += blockDim.x) {
+    const long off = p << 4;
+    uint32_t acc[NOUT][4];
+#pragma unroll
+    for (int j = 0; j < NOUT; j++)
+      acc[j][0] = acc[j][1] = acc[j][2] = acc[j][3] = 0u;
+
+    for (int i = 0; i < n_src; i++) {
+      const uint8_t* sp = (const uint8_t*)ptrs[s_src[i]];
+      if (sp == nullptr) continue;  // zeros chunk for this slice
+      const v4u* p4 = reinterpret_cast<const v4u*>(sp + off);
+      const v4u d = NT ? __builtin_nontemporal_load(p4) : *p4;
+      const uint32_t dq[4] = {d.x, d.y, d.z, d.w};
+      uint32_t i7l[4], i8l[4], i7h[4], i8h[4];
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+        i7l[q] = dq[q] & 0x07070707u;
+        i8l[q] = (dq[q] >> 3) & 0x01010101u;
+        i7h[q] = (dq[q] >> 4) & 0x07070707u;
+        i8h[q] = (dq[q] >> 7) & 0x01010101u;
+      }
+#pragma unroll
+      for (int j = 0; j < NOUT; j++) {
+        const int cls = __builtin_amdgcn_readfirstlane(s_cls[j * n_src + i]);
+        if (cls == 0) continue;
+        if (cls == 1) {
+#pragma unroll
+          for (int q = 0; q < 4; q++) acc[j][q] ^= dq[q];
+        } else {
+          const uint32_t* T = &s_tabs[(j * n_src + i) * 6];
+          const uint32_t t0 = T[0], t1 = T[1], t2 = T[2], t3 = T[3],
+                         t4 = T[4], t5 = T[5];
+#pragma unroll
+          for (int q = 0; q < 4; q++)
+            acc[j][q] ^= ecx_lut(t0, t1, t2, i7l[q], i8l[q]) ^
+                         ecx_lut(t3, t4, t5, i7h[q], i8h[q]);
+        }
+      }
+    }
+
+#pragma unroll
+    for (int j = 0; j < NOUT; j++) {
+      v4u o;
+      o.x = acc[j][0]; o.y = acc[j][1]; o.z = acc[j][2]; o.w = acc[j][3];
+      v4u* p4 = reinterpret_cast<v4u*>((uint8_t*)ptrs[s_out[j]] + off);
+      if (NT)
+        __builtin_nontemporal_store(o, p4);
+      else
+        *p4 = o;
+    }
+  }
+}
+
 // delta = a ^ b (encode_delta; replaces galois_region_xor / xor_gen).
 __global__ __launch_bounds__(256) void ec_xor_kernel(
     const uint8_t* __restrict__ a, const uint8_t* __restrict__ b,
@@ -234,6 +320,11 @@ struct Slot {
   hipEvent_t ev_param = nullptr;  // param upload completion (ring of 1)
   EcLaunchParams* h_params = nullptr;  // pinned
   EcLaunchParams* d_params = nullptr;
+  // variable-size slice batch scratch (job table), grown on demand
+  hipEvent_t ev_jobs = nullptr;
+  uint8_t* h_jobs = nullptr;  // pinned
+  uint8_t* d_jobs = nullptr;
+  size_t jobs_bytes = 0;
   // host-path staging stripe buffer, grown on demand
   uint8_t* d_stage = nullptr;
   size_t stage_bytes = 0;
@@ -314,6 +405,7 @@ int ecx_create(int k, int m, int technique, int device, int n_streams,
         hipEventCreate(&s.ev_start) != hipSuccess ||
         hipEventCreate(&s.ev_stop) != hipSuccess ||
         hipEventCreate(&s.ev_param) != hipSuccess ||
+        hipEventCreate(&s.ev_jobs) != hipSuccess ||
         hipHostMalloc(&s.h_params, sizeof(EcLaunchParams)) != hipSuccess ||
         hipMalloc(&s.d_params, sizeof(EcLaunchParams)) != hipSuccess) {
       ecx_destroy(ctx);
@@ -332,6 +424,9 @@ void ecx_destroy(ecx_ctx* ctx) {
     if (s.ev_start) (void)hipEventDestroy(s.ev_start);
     if (s.ev_stop) (void)hipEventDestroy(s.ev_stop);
     if (s.ev_param) (void)hipEventDestroy(s.ev_param);
+    if (s.ev_jobs) (void)hipEventDestroy(s.ev_jobs);
+    if (s.h_jobs) (void)hipHostFree(s.h_jobs);
+    if (s.d_jobs) (void)hipFree(s.d_jobs);
     if (s.h_params) (void)hipHostFree(s.h_params);
     if (s.d_params) (void)hipFree(s.d_params);
     if (s.d_stage) (void)hipFree(s.d_stage);
@@ -503,9 +598,12 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
     int x = v ? atoi(v) : 4;
     return x >= 1 ? x : 4;
   }();
+  // NT=1 (nontemporal loads/stores) measured +7% encode bandwidth —
+  // streaming data with zero reuse should not occupy L1/L2
+  // (profiles/rocprof_r01_summary.md).
   static const int env_nt = [] {
     const char* v = getenv("ECX_NT");
-    return v ? atoi(v) : 0;
+    return v ? atoi(v) : 1;
   }();
   const int vpt = (vecs >= 2 * 256) ? env_vpt : 1;
   const long per_block = 256L * vpt;
@@ -581,6 +679,122 @@ static int run_matmul(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
     int r = launch_matmul(ctx, s, d_buf, d_obuf, p, n_stripes, chunk_bytes,
                           accum, j0 == 0);
     if (r != ECX_OK) return r;
+  }
+  return ECX_OK;
+}
+
+static int ensure_jobs(ecx_ctx* ctx, Slot& s, size_t bytes) {
+  if (s.jobs_bytes >= bytes) return ECX_OK;
+  HIP_TRY(hipSetDevice(ctx->device));
+  HIP_TRY(hipEventSynchronize(s.ev_jobs));
+  if (s.h_jobs) (void)hipHostFree(s.h_jobs);
+  if (s.d_jobs) (void)hipFree(s.d_jobs);
+  s.h_jobs = nullptr;
+  s.d_jobs = nullptr;
+  s.jobs_bytes = 0;
+  size_t want = bytes + bytes / 2 + 4096;  // grow with headroom
+  HIP_TRY(hipHostMalloc(&s.h_jobs, want));
+  HIP_TRY(hipMalloc(&s.d_jobs, want));
+  s.jobs_bytes = want;
+  return ECX_OK;
+}
+
+// Launch the variable-size slice kernel for one <=4-output group.
+static int run_slices(ecx_ctx* ctx, int slot_i, void* const* d_chunks,
+                      const size_t* bytes, int n_slices, const int* src_ids,
+                      int n_src, const int* out_ids, int n_out,
+                      const uint8_t* coeff) {
+  if (!ctx || slot_i < 0 || slot_i >= (int)ctx->slots.size() ||
+      n_slices < 1 || n_src < 1 || n_src > ECX_MAX_K || n_out < 1)
+    return ECX_ERR_INVAL;
+  const int cps = ctx->k + ctx->m;
+  for (int i = 0; i < n_slices; i++)
+    if (bytes[i] == 0 || bytes[i] % 16) return ECX_ERR_INVAL;
+  Slot& s = ctx->slots[slot_i];
+  std::lock_guard<std::recursive_mutex> g(s.mu);
+  HIP_TRY(hipSetDevice(ctx->device));
+
+  static const int env_nt = [] {
+    const char* v = getenv("ECX_NT");
+    return v ? atoi(v) : 1;
+  }();
+  const int VECS_PER_BLOCK = 256 * 4;
+
+  // block assignment + blob layout: [ptrs u64][slice_vecs i64]
+  // [block_voff i64][block_slice i32]
+  long n_blocks = 0;
+  for (int i = 0; i < n_slices; i++)
+    n_blocks += (long)((bytes[i] >> 4) + VECS_PER_BLOCK - 1) / VECS_PER_BLOCK;
+  if (n_blocks > (1 << 30)) return ECX_ERR_INVAL;
+  size_t off_ptrs = 0;
+  size_t off_vecs = off_ptrs + (size_t)n_slices * cps * 8;
+  size_t off_voff = off_vecs + (size_t)n_slices * 8;
+  size_t off_bsl = off_voff + (size_t)n_blocks * 8;
+  size_t blob = off_bsl + (size_t)n_blocks * 4;
+  int r = ensure_jobs(ctx, s, blob);
+  if (r != ECX_OK) return r;
+  HIP_TRY(hipEventSynchronize(s.ev_jobs));
+
+  uint64_t* ptrs = (uint64_t*)(s.h_jobs + off_ptrs);
+  long* svecs = (long*)(s.h_jobs + off_vecs);
+  long* bvoff = (long*)(s.h_jobs + off_voff);
+  int* bslice = (int*)(s.h_jobs + off_bsl);
+  long b = 0;
+  for (int i = 0; i < n_slices; i++) {
+    for (int c = 0; c < cps; c++)
+      ptrs[(size_t)i * cps + c] = (uint64_t)d_chunks[(size_t)i * cps + c];
+    long nv = (long)(bytes[i] >> 4);
+    svecs[i] = nv;
+    for (long v = 0; v < nv; v += VECS_PER_BLOCK) {
+      bslice[b] = i;
+      bvoff[b] = v;
+      b++;
+    }
+  }
+  HIP_TRY(hipMemcpyAsync(s.d_jobs, s.h_jobs, blob, hipMemcpyHostToDevice,
+                         s.stream));
+  HIP_TRY(hipEventRecord(s.ev_jobs, s.stream));
+
+  const uint64_t* d_ptrs = (const uint64_t*)(s.d_jobs + off_ptrs);
+  const long* d_svecs = (const long*)(s.d_jobs + off_vecs);
+  const long* d_bvoff = (const long*)(s.d_jobs + off_voff);
+  const int* d_bslice = (const int*)(s.d_jobs + off_bsl);
+
+  const ecx::GF8& f = ecx::gf8();
+  for (int j0 = 0; j0 < n_out; j0 += 4) {
+    int nj = std::min(4, n_out - j0);
+    EcLaunchParams p;
+    std::vector<uint8_t> sub((size_t)nj * n_src);
+    for (int j = 0; j < nj; j++)
+      std::memcpy(&sub[(size_t)j * n_src], &coeff[(size_t)(j0 + j) * n_src],
+                  n_src);
+    // per-slice NULL chunks are handled inside the kernel, not via cls
+    fill_params(&p, f, src_ids, n_src, out_ids + j0, nj, sub.data(), nullptr);
+    HIP_TRY(hipEventSynchronize(s.ev_param));
+    std::memcpy(s.h_params, &p, sizeof(EcLaunchParams));
+    HIP_TRY(hipMemcpyAsync(s.d_params, s.h_params, sizeof(EcLaunchParams),
+                           hipMemcpyHostToDevice, s.stream));
+    HIP_TRY(hipEventRecord(s.ev_param, s.stream));
+#define ECX_SLAUNCH(NO, NTF)                                                \
+  hipLaunchKernelGGL((ec_gf_slices_kernel<NO, NTF>), dim3((unsigned)n_blocks), \
+                     dim3(256), 0, s.stream, d_ptrs, d_bslice, d_bvoff,     \
+                     d_svecs, s.d_params, cps, VECS_PER_BLOCK)
+#define ECX_SDISPATCH(NO)                  \
+  case NO:                                 \
+    if (env_nt) ECX_SLAUNCH(NO, true);     \
+    else ECX_SLAUNCH(NO, false);           \
+    break;
+    switch (nj) {
+      ECX_SDISPATCH(1)
+      ECX_SDISPATCH(2)
+      ECX_SDISPATCH(3)
+      ECX_SDISPATCH(4)
+      default:
+        return ECX_ERR_INVAL;
+    }
+#undef ECX_SDISPATCH
+#undef ECX_SLAUNCH
+    HIP_TRY(hipGetLastError());
   }
   return ECX_OK;
 }
@@ -670,6 +884,30 @@ int ecx_apply_delta_dev(ecx_ctx* ctx, const void* d_delta, int data_shard,
   // chunk_bytes = bytes, chunks_per_stripe irrelevant with single stripe.
   return run_matmul(ctx, slot, (const uint8_t*)d_delta, (uint8_t*)d_parity,
                     src_ids, 1, out_ids, 1, coeff, nullptr, 1, bytes, true);
+}
+
+int ecx_encode_slices(ecx_ctx* ctx, void* const* d_chunks,
+                      const size_t* bytes, int n_slices, int slot) {
+  if (!ctx || !d_chunks || !bytes) return ECX_ERR_INVAL;
+  int k = ctx->k, m = ctx->m;
+  int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
+  for (int i = 0; i < k; i++) src_ids[i] = i;
+  for (int j = 0; j < m; j++) out_ids[j] = k + j;
+  return run_slices(ctx, slot, d_chunks, bytes, n_slices, src_ids, k,
+                    out_ids, m, ctx->gen.data() + (size_t)k * k);
+}
+
+int ecx_decode_slices(ecx_ctx* ctx, void* const* d_chunks,
+                      const size_t* bytes, int n_slices,
+                      uint64_t present_mask, int slot) {
+  if (!ctx || !d_chunks || !bytes) return ECX_ERR_INVAL;
+  DecodePlan plan;
+  int r = get_decode_plan(ctx, present_mask, plan);
+  if (r != ECX_OK) return r;
+  if (plan.erased.empty()) return ECX_OK;
+  return run_slices(ctx, slot, d_chunks, bytes, n_slices,
+                    plan.survivors.data(), ctx->k, plan.erased.data(),
+                    (int)plan.erased.size(), plan.rows.data());
 }
 
 int ecx_sync(ecx_ctx* ctx, int slot) {

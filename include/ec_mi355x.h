@@ -156,6 +156,25 @@ ECX_API int ecx_apply_delta_dev(ecx_ctx *ctx, const void *d_delta,
                                 int data_shard, int coding_shard,
                                 void *d_parity, size_t bytes, int slot);
 
+/* ---------------- variable-size slice batch (caller-shaped) --
+ * The OSD write path calls encode_chunks once per page-aligned slice with
+ * varying blocksize (shard_extent_map_t::encode, src/osd/ECUtil.cc:485-514,
+ * EC_ALIGN_SIZE=4096) — many small calls. This entry point batches N such
+ * slices into ONE kernel launch over device-resident chunk pointers.
+ * Arrays are host-side, length n_slices:
+ *   d_chunks[s*(k+m)+c] = device pointer to chunk c of slice s (data
+ *     0..k-1 may be NULL => zeros);
+ *   bytes[s] = slice length (multiple of 16).
+ * Encode: parity pointers k..k+m-1 are written.
+ */
+ECX_API int ecx_encode_slices(ecx_ctx *ctx, void *const *d_chunks,
+                              const size_t *bytes, int n_slices, int slot);
+
+/* Same for decode under a per-call uniform erasure pattern. */
+ECX_API int ecx_decode_slices(ecx_ctx *ctx, void *const *d_chunks,
+                              const size_t *bytes, int n_slices,
+                              uint64_t present_mask, int slot);
+
 /* Synchronise a stream slot. */
 ECX_API int ecx_sync(ecx_ctx *ctx, int slot);
 
