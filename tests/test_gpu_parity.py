@@ -207,3 +207,36 @@ def test_decode_lru_reuse():
                 assert np.array_equal(chunks[i], full[i]), (trial, i)
     finally:
         ctx.close()
+
+
+def test_concurrent_calls_one_ctx():
+    """One codec instance must be safe from many threads (the reference's
+    plugins are called from many PG threads; ErasureCodeInterface.h
+    threading contract — our slots serialize internally)."""
+    import threading
+    k, m = 8, 3
+    C = 256 * 1024
+    tech = "reed_sol_van"
+    rng = np.random.default_rng(33)
+    ctx = make_ctx(k, m, tech)
+    datasets = [[rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+                for _ in range(4)]
+    wants = [oracle.encode(tech, k, m, d) for d in datasets]
+    errors = []
+
+    def worker(t):
+        try:
+            for _ in range(5):
+                got = ctx.encode_chunks(datasets[t])
+                for j in range(m):
+                    assert np.array_equal(got[j], wants[t][j])
+        except Exception as e:  # pragma: no cover
+            errors.append((t, repr(e)))
+
+    threads = [threading.Thread(target=worker, args=(t,)) for t in range(4)]
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join()
+    ctx.close()
+    assert not errors, errors
