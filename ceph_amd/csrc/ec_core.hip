@@ -585,9 +585,9 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
   const long vecs = (long)(chunk_bytes >> 4);
   // Tunables (A/B-able via env on the GPU box): VPT = 16B vectors per
   // thread per iteration, TILE = target vectors per thread per launch.
-  // Defaults from the round-1 sweep (profiles/rocprof_r01_summary.md):
-  // VPT=1 keeps 8 waves/SIMD at NOUT=3 and beat VPT=2; TILE=4 won the
-  // grid-shape sweep.
+  // Defaults from the round-1 sweeps (profiles/rocprof_r01_summary.md):
+  // VPT=1 keeps 8 waves/SIMD at NOUT=3 and beat VPT=2; with NT on,
+  // TILE=2 won the grid-shape sweep (6.0 TB/s encode).
   static const int env_vpt = [] {
     const char* v = getenv("ECX_VPT");
     int x = v ? atoi(v) : 1;
@@ -595,8 +595,8 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
   }();
   static const int env_tile = [] {
     const char* v = getenv("ECX_TILE");
-    int x = v ? atoi(v) : 4;
-    return x >= 1 ? x : 4;
+    int x = v ? atoi(v) : 2;
+    return x >= 1 ? x : 2;
   }();
   // NT=1 (nontemporal loads/stores) measured +7% encode bandwidth —
   // streaming data with zero reuse should not occupy L1/L2
