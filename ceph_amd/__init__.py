@@ -16,10 +16,12 @@ _SO = os.path.join(_DIR, "libec_mi355x_core.so")
 T_RS_VAN_ISA = 0
 T_CAUCHY_ISA = 1
 T_RS_VAN_JERASURE = 2
+T_CAUCHY_ORIG_JERASURE = 3
 TECHNIQUES = {
     "reed_sol_van": T_RS_VAN_ISA,
     "cauchy": T_CAUCHY_ISA,
     "jerasure_reed_sol_van": T_RS_VAN_JERASURE,
+    "cauchy_orig": T_CAUCHY_ORIG_JERASURE,
 }
 
 _ERR = {
@@ -42,6 +44,7 @@ def _lib():
     lib.ecx_version.restype = ctypes.c_char_p
     lib.ecx_device_count.restype = ctypes.c_int
     lib.ecx_create.argtypes = [ctypes.c_int] * 5 + [ctypes.POINTER(ctypes.c_void_p)]
+    lib.ecx_create2.argtypes = [ctypes.c_int] * 7 + [ctypes.POINTER(ctypes.c_void_p)]
     lib.ecx_chunk_size.restype = ctypes.c_uint
     lib.ecx_chunk_size.argtypes = [ctypes.c_void_p, ctypes.c_uint]
     lib.ecx_minimum_to_decode.argtypes = [
@@ -131,12 +134,14 @@ class EcContext:
     """One (k, m, technique) codec bound to one GPU, mirroring a plugin
     instance after init()/prepare() (ErasureCodeIsa.cc:637-697)."""
 
-    def __init__(self, k, m, technique="reed_sol_van", device=0, n_streams=2):
+    def __init__(self, k, m, technique="reed_sol_van", device=0,
+                 n_streams=2, packetsize=2048, w=8):
         t = TECHNIQUES[technique] if isinstance(technique, str) else technique
         self._h = ctypes.c_void_p()
         self.k, self.m, self.technique = k, m, technique
-        _ck(lib().ecx_create(k, m, t, device, n_streams,
-                             ctypes.byref(self._h)), "ecx_create")
+        self.packetsize = packetsize
+        _ck(lib().ecx_create2(k, m, t, w, packetsize, device, n_streams,
+                              ctypes.byref(self._h)), "ecx_create2")
 
     def close(self):
         if self._h:

@@ -277,6 +277,83 @@ __global__ __launch_bounds__(256, 2) void ec_gf_slices_kernel(
   }
 }
 
+// ---- jerasure bitmatrix (Cauchy-original) path ----
+// Packet-sliced XOR gather: each chunk is a stream of superwords (w packets
+// of `pkt` bytes); coding packet row r = XOR of the data packets its bit
+// row selects (companion-basis GF(2^8), see gf.cpp matrix_to_bitmatrix).
+// A naive gather would re-read each data packet ~m*w/2 times from HBM; this
+// kernel stages a q-byte window of all k*w packets in LDS once and computes
+// every output row from LDS, keeping HBM traffic algorithmic ((k+m)/k) —
+// the CDNA-native answer to jerasure's CPU XOR schedules.
+struct EcBitParams {
+  int n_src;   // k
+  int n_out;   // output chunks this launch (rows = n_out*w)
+  int w;
+  int pkt;     // packetsize bytes
+  int q;       // LDS window bytes (power-of-two divisor of pkt)
+  int vq_shift;  // log2(q/16)
+  int src_ids[ECX_MAX_K];
+  int out_ids[ECX_MAX_OUT];
+  uint16_t row_off[ECX_MAX_OUT * 8 + 1];  // prefix offsets into ops[]
+  // blob continues with uint16 ops[row_off[n_rows]]: values j*w+c
+};
+
+template <bool NT>
+__global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
+    const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
+    const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
+    int windows_per_sw) {
+  const EcBitParams* bp = (const EcBitParams*)blob;
+  const uint16_t* g_ops = (const uint16_t*)(blob + sizeof(EcBitParams));
+  extern __shared__ uint8_t smem[];
+  const int n_src = bp->n_src, w = bp->w, pkt = bp->pkt, q = bp->q;
+  const int vq = q >> 4, vq_shift = bp->vq_shift;
+  const int n_rows = bp->n_out * w;
+  uint8_t* s_data = smem;                       // n_src*w*q bytes
+  uint16_t* s_ops = (uint16_t*)(smem + (size_t)n_src * w * q);
+  const int n_ops = bp->row_off[n_rows];
+  for (int t = threadIdx.x; t < n_ops; t += blockDim.x) s_ops[t] = g_ops[t];
+
+  const long tile = blockIdx.x;
+  const int win = (int)(tile % windows_per_sw);
+  const long sw = tile / windows_per_sw;
+  const uint8_t* sbase = buf + (long)blockIdx.y * cps * chunk_bytes;
+  uint8_t* obase = obuf + (long)blockIdx.y * cps * chunk_bytes;
+  const long sw_off = sw * (long)w * pkt + (long)win * q;
+
+  for (int t = threadIdx.x; t < n_src * w * vq; t += blockDim.x) {
+    const int jc = t >> vq_shift;
+    const int v = t - (jc << vq_shift);
+    const int j = jc / w, c = jc - j * w;
+    const v4u* src = reinterpret_cast<const v4u*>(
+        sbase + (long)bp->src_ids[j] * chunk_bytes + sw_off + (long)c * pkt +
+        (long)v * 16);
+    const v4u d = NT ? __builtin_nontemporal_load(src) : *src;
+    *reinterpret_cast<v4u*>(s_data + (size_t)jc * q + (size_t)v * 16) = d;
+  }
+  __syncthreads();
+
+  for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
+    const int r = t >> vq_shift;
+    const int v = t - (r << vq_shift);
+    v4u acc = {0, 0, 0, 0};
+    const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
+    for (int o = b0; o < b1; o++) {
+      const int jc = s_ops[o];
+      const v4u d =
+          *reinterpret_cast<const v4u*>(s_data + (size_t)jc * q + (size_t)v * 16);
+      acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
+    }
+    v4u* dst = reinterpret_cast<v4u*>(
+        obase + (long)bp->out_ids[r / w] * chunk_bytes + sw * (long)w * pkt +
+        (long)(r % w) * pkt + (long)win * q + (long)v * 16);
+    if (NT)
+      __builtin_nontemporal_store(acc, dst);
+    else
+      *dst = acc;
+  }
+}
+
 // delta = a ^ b (encode_delta; replaces galois_region_xor / xor_gen).
 __global__ __launch_bounds__(256) void ec_xor_kernel(
     const uint8_t* __restrict__ a, const uint8_t* __restrict__ b,
@@ -339,11 +416,19 @@ struct DecodePlan {
   std::vector<uint8_t> rows;  // n_erased x k
 };
 
+struct BitPlan {
+  std::vector<int> survivors;
+  std::vector<int> erased;
+  std::vector<uint8_t> rows;  // (n_erased*w) x (k*w) bits
+};
+
 }  // namespace
 
 struct ecx_ctx {
   int k = 0, m = 0, technique = 0, device = 0;
-  std::vector<uint8_t> gen;  // (k+m) x k
+  int w = 8, pkt = 2048;  // bitmatrix techniques only (jerasure packetsize)
+  std::vector<uint8_t> gen;     // (k+m) x k
+  std::vector<uint8_t> bitmat;  // (m*w) x (k*w) bits (bitmatrix techniques)
   std::vector<Slot> slots;
   // decode-plan LRU keyed by present_mask (exact signature for fixed
   // (k,m,technique) — the analogue of ErasureCodeIsaTableCache's
@@ -352,7 +437,10 @@ struct ecx_ctx {
   std::mutex lru_mu;
   std::map<uint64_t, std::pair<DecodePlan, std::list<uint64_t>::iterator>> lru;
   std::list<uint64_t> lru_order;
+  std::map<uint64_t, BitPlan> bit_lru;  // bitmatrix decode plans
   static constexpr size_t LRU_DEPTH = 4096;
+
+  bool is_bitmatrix() const { return technique == ECX_T_CAUCHY_ORIG_JERASURE; }
 };
 
 static int map_hip(hipError_t e) {
@@ -378,10 +466,13 @@ int ecx_device_count(void) {
   return n;
 }
 
-int ecx_create(int k, int m, int technique, int device, int n_streams,
-               ecx_ctx** out) {
+int ecx_create2(int k, int m, int technique, int w, int packetsize,
+                int device, int n_streams, ecx_ctx** out) {
   if (!out || k < 2 || m < 1 || k > ECX_MAX_K || m > ECX_MAX_K ||
-      n_streams < 1 || n_streams > 64)
+      n_streams < 1 || n_streams > 64 || w != 8)
+    return ECX_ERR_INVAL;
+  if (technique == ECX_T_CAUCHY_ORIG_JERASURE &&
+      (packetsize < 16 || packetsize % 16 || k > 16))
     return ECX_ERR_INVAL;
   if (ecx_device_count() <= device) return ECX_ERR_NO_GPU;
 
@@ -390,10 +481,15 @@ int ecx_create(int k, int m, int technique, int device, int n_streams,
   ctx->m = m;
   ctx->technique = technique;
   ctx->device = device;
+  ctx->w = w;
+  ctx->pkt = packetsize;
   if (!ecx::gen_matrix(technique, ctx->gen, k, m)) {
     delete ctx;
     return ECX_ERR_INVAL;
   }
+  if (ctx->is_bitmatrix())
+    ecx::matrix_to_bitmatrix(ctx->gen.data() + (size_t)k * k, k, m, w,
+                             ctx->bitmat);
   hipError_t e = hipSetDevice(device);
   if (e != hipSuccess) {
     delete ctx;
@@ -414,6 +510,11 @@ int ecx_create(int k, int m, int technique, int device, int n_streams,
   }
   *out = ctx;
   return ECX_OK;
+}
+
+int ecx_create(int k, int m, int technique, int device, int n_streams,
+               ecx_ctx** out) {
+  return ecx_create2(k, m, technique, 8, 2048, device, n_streams, out);
 }
 
 void ecx_destroy(ecx_ctx* ctx) {
@@ -446,6 +547,15 @@ int ecx_get_matrix(const ecx_ctx* ctx, uint8_t* out) {
 
 unsigned ecx_chunk_size(const ecx_ctx* ctx, unsigned stripe_width) {
   if (!ctx) return 0;
+  if (ctx->is_bitmatrix()) {
+    // ErasureCodeJerasureCauchy::get_alignment (per_chunk_alignment=false):
+    // stripe aligned to k*w*packetsize*sizeof(int) => chunk is a multiple
+    // of w*packetsize*4 (ErasureCodeJerasure.cc:522-536)
+    unsigned align = (unsigned)ctx->k * ctx->w * ctx->pkt * 4u;
+    unsigned tail = stripe_width % align;
+    unsigned padded = stripe_width + (tail ? align - tail : 0);
+    return padded / ctx->k;
+  }
   if (ctx->technique == ECX_T_RS_VAN_JERASURE) {
     // ErasureCodeJerasure.cc:85-108, w=8, per_chunk_alignment=false
     unsigned align = (unsigned)ctx->k * 8u * 4u;
@@ -799,6 +909,108 @@ static int run_slices(ecx_ctx* ctx, int slot_i, void* const* d_chunks,
   return ECX_OK;
 }
 
+// Launch the bitmatrix kernel for <= ECX_MAX_OUT output chunks whose bit
+// rows (n_out*w x n_src*w) are given over the source chunk ids.
+static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
+                         uint8_t* d_obuf, const int* src_ids, int n_src,
+                         const int* out_ids, int n_out,
+                         const uint8_t* bit_rows, long n_stripes,
+                         size_t chunk_bytes, bool time_it) {
+  const int w = ctx->w, pkt = ctx->pkt;
+  if (n_src < 1 || n_src > 16 || n_out < 1 || n_out > ECX_MAX_OUT)
+    return ECX_ERR_INVAL;
+  if (chunk_bytes == 0 || chunk_bytes % ((size_t)w * pkt) || n_stripes <= 0 ||
+      n_stripes > 65535)
+    return ECX_ERR_INVAL;
+  Slot& s = ctx->slots[slot_i];
+  std::lock_guard<std::recursive_mutex> g(s.mu);
+  HIP_TRY(hipSetDevice(ctx->device));
+
+  static const int env_nt = [] {
+    const char* v = getenv("ECX_NT");
+    return v ? atoi(v) : 1;
+  }();
+
+  // LDS window: largest power-of-two divisor of pkt with n_src*w*q <= 48 KB
+  int q = 16;
+  while (q * 2 <= pkt && pkt % (q * 2) == 0 &&
+         (size_t)n_src * w * q * 2 <= 48 * 1024)
+    q *= 2;
+  int vq_shift = 0;
+  while ((1 << vq_shift) < q / 16) vq_shift++;
+  if ((16 << vq_shift) != q) return ECX_ERR_INVAL;
+
+  // build blob: header + ops
+  const int n_rows = n_out * w, W = n_src * w;
+  EcBitParams hdr;
+  std::memset(&hdr, 0, sizeof(hdr));
+  hdr.n_src = n_src;
+  hdr.n_out = n_out;
+  hdr.w = w;
+  hdr.pkt = pkt;
+  hdr.q = q;
+  hdr.vq_shift = vq_shift;
+  for (int i = 0; i < n_src; i++) hdr.src_ids[i] = src_ids[i];
+  for (int j = 0; j < n_out; j++) hdr.out_ids[j] = out_ids[j];
+  std::vector<uint16_t> ops;
+  ops.reserve((size_t)n_rows * W / 2);
+  for (int r = 0; r < n_rows; r++) {
+    hdr.row_off[r] = (uint16_t)ops.size();
+    const uint8_t* row = bit_rows + (size_t)r * W;
+    for (int c = 0; c < W; c++)
+      if (row[c]) ops.push_back((uint16_t)c);
+  }
+  hdr.row_off[n_rows] = (uint16_t)ops.size();
+  if (ops.size() > 0xffff) return ECX_ERR_INVAL;
+  size_t blob = sizeof(EcBitParams) + ops.size() * 2;
+  int r = ensure_jobs(ctx, s, blob);
+  if (r != ECX_OK) return r;
+  HIP_TRY(hipEventSynchronize(s.ev_jobs));
+  std::memcpy(s.h_jobs, &hdr, sizeof(hdr));
+  std::memcpy(s.h_jobs + sizeof(hdr), ops.data(), ops.size() * 2);
+  HIP_TRY(hipMemcpyAsync(s.d_jobs, s.h_jobs, blob, hipMemcpyHostToDevice,
+                         s.stream));
+  HIP_TRY(hipEventRecord(s.ev_jobs, s.stream));
+
+  const long sw_per_chunk = (long)(chunk_bytes / ((size_t)w * pkt));
+  const int windows_per_sw = pkt / q;
+  dim3 grid((unsigned)(sw_per_chunk * windows_per_sw), (unsigned)n_stripes);
+  size_t lds = (size_t)n_src * w * q + ((ops.size() * 2 + 15) & ~15ull);
+  if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
+  if (env_nt)
+    hipLaunchKernelGGL((ec_bitmatrix_kernel<true>), grid, dim3(256), lds,
+                       s.stream, d_buf, d_obuf, s.d_jobs, (long)chunk_bytes,
+                       ctx->k + ctx->m, windows_per_sw);
+  else
+    hipLaunchKernelGGL((ec_bitmatrix_kernel<false>), grid, dim3(256), lds,
+                       s.stream, d_buf, d_obuf, s.d_jobs, (long)chunk_bytes,
+                       ctx->k + ctx->m, windows_per_sw);
+  HIP_TRY(hipGetLastError());
+  if (time_it) {
+    HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
+    s.timed = true;
+  }
+  return ECX_OK;
+}
+
+static int get_bit_plan(ecx_ctx* ctx, uint64_t present_mask, BitPlan& out) {
+  std::lock_guard<std::mutex> g(ctx->lru_mu);
+  auto it = ctx->bit_lru.find(present_mask);
+  if (it != ctx->bit_lru.end()) {
+    out = it->second;
+    return ECX_OK;
+  }
+  BitPlan plan;
+  if (!ecx::compose_bit_decode_rows(ctx->bitmat, ctx->k, ctx->m, ctx->w,
+                                    present_mask, plan.survivors,
+                                    plan.erased, plan.rows))
+    return ECX_ERR_IO;
+  if (ctx->bit_lru.size() > ecx_ctx::LRU_DEPTH) ctx->bit_lru.clear();
+  ctx->bit_lru.emplace(present_mask, plan);
+  out = plan;
+  return ECX_OK;
+}
+
 static int get_decode_plan(ecx_ctx* ctx, uint64_t present_mask,
                            DecodePlan& out) {
   std::lock_guard<std::mutex> g(ctx->lru_mu);
@@ -834,6 +1046,12 @@ int ecx_encode_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
   int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
   for (int i = 0; i < k; i++) src_ids[i] = i;
   for (int j = 0; j < m; j++) out_ids[j] = k + j;
+  if (ctx->is_bitmatrix()) {
+    if (slot < 0 || slot >= (int)ctx->slots.size()) return ECX_ERR_INVAL;
+    return run_bitmatrix(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr,
+                         src_ids, k, out_ids, m, ctx->bitmat.data(),
+                         n_stripes, chunk_bytes, true);
+  }
   const uint8_t* rows = ctx->gen.data() + (size_t)k * k;
   return run_matmul(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr, src_ids,
                     k, out_ids, m, rows, nullptr, n_stripes, chunk_bytes,
@@ -843,6 +1061,17 @@ int ecx_encode_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
 int ecx_decode_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
                      size_t chunk_bytes, uint64_t present_mask, int slot) {
   if (!ctx || !dptr) return ECX_ERR_INVAL;
+  if (ctx->is_bitmatrix()) {
+    if (slot < 0 || slot >= (int)ctx->slots.size()) return ECX_ERR_INVAL;
+    BitPlan plan;
+    int r = get_bit_plan(ctx, present_mask, plan);
+    if (r != ECX_OK) return r;
+    if (plan.erased.empty()) return ECX_OK;
+    return run_bitmatrix(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr,
+                         plan.survivors.data(), ctx->k, plan.erased.data(),
+                         (int)plan.erased.size(), plan.rows.data(),
+                         n_stripes, chunk_bytes, false);
+  }
   DecodePlan plan;
   int r = get_decode_plan(ctx, present_mask, plan);
   if (r != ECX_OK) return r;
@@ -967,6 +1196,21 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
   int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
   for (int i = 0; i < k; i++) src_ids[i] = i;
   for (int j = 0; j < m; j++) out_ids[j] = k + j;
+  if (ctx->is_bitmatrix()) {
+    // zeros-chunk convention: materialise zeros in the stage buffer
+    for (int i = 0; i < k; i++)
+      if (src_null[i])
+        HIP_TRY(hipMemsetAsync(s.d_stage + (size_t)i * chunk_bytes, 0,
+                               chunk_bytes, s.stream));
+    for (int j0 = 0; j0 < m; j0 += ECX_MAX_OUT) {
+      int nj = std::min(ECX_MAX_OUT, m - j0);
+      int rr = run_bitmatrix(
+          ctx, 0, s.d_stage, s.d_stage, src_ids, k, out_ids + j0, nj,
+          ctx->bitmat.data() + (size_t)(j0 * ctx->w) * k * ctx->w, 1,
+          chunk_bytes, false);
+      if (rr != ECX_OK) return rr;
+    }
+  } else {
   const uint8_t* rows = ctx->gen.data() + (size_t)k * k;
   const ecx::GF8& f = ecx::gf8();
   for (int j0 = 0; j0 < m; j0 += 4) {
@@ -979,6 +1223,7 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
     int rr = launch_matmul(ctx, s, s.d_stage, s.d_stage, p, 1, chunk_bytes,
                            false, false);
     if (rr != ECX_OK) return rr;
+  }
   }
   for (int j = 0; j < m; j++) {
     if (!parity[j]) continue;
@@ -993,6 +1238,46 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
                            uint64_t present_mask, size_t chunk_bytes) {
   if (!ctx || !chunks || chunk_bytes % 16) return ECX_ERR_INVAL;
   int k = ctx->k, m = ctx->m, n = k + m;
+
+  if (ctx->is_bitmatrix()) {
+    BitPlan plan;
+    int r = get_bit_plan(ctx, present_mask, plan);
+    if (r != ECX_OK) return r;
+    if (plan.erased.empty()) return ECX_OK;
+    Slot& s = ctx->slots[0];
+    std::lock_guard<std::recursive_mutex> g(s.mu);
+    r = ensure_stage(ctx, s, (size_t)n * chunk_bytes);
+    if (r != ECX_OK) return r;
+    HIP_TRY(hipSetDevice(ctx->device));
+    for (int i = 0; i < k; i++) {
+      int id = plan.survivors[i];
+      if (!chunks[id]) {
+        HIP_TRY(hipMemsetAsync(s.d_stage + (size_t)id * chunk_bytes, 0,
+                               chunk_bytes, s.stream));
+        continue;
+      }
+      HIP_TRY(hipMemcpyAsync(s.d_stage + (size_t)id * chunk_bytes,
+                             chunks[id], chunk_bytes, hipMemcpyHostToDevice,
+                             s.stream));
+    }
+    for (size_t j0 = 0; j0 < plan.erased.size(); j0 += ECX_MAX_OUT) {
+      int nj = (int)std::min<size_t>(ECX_MAX_OUT, plan.erased.size() - j0);
+      int rr = run_bitmatrix(
+          ctx, 0, s.d_stage, s.d_stage, plan.survivors.data(), k,
+          plan.erased.data() + j0, nj,
+          plan.rows.data() + j0 * (size_t)ctx->w * k * ctx->w, 1,
+          chunk_bytes, false);
+      if (rr != ECX_OK) return rr;
+    }
+    for (int e : plan.erased) {
+      if (!chunks[e]) return ECX_ERR_INVAL;
+      HIP_TRY(hipMemcpyAsync(chunks[e], s.d_stage + (size_t)e * chunk_bytes,
+                             chunk_bytes, hipMemcpyDeviceToHost, s.stream));
+    }
+    HIP_TRY(hipStreamSynchronize(s.stream));
+    return ECX_OK;
+  }
+
   DecodePlan plan;
   int r = get_decode_plan(ctx, present_mask, plan);
   if (r != ECX_OK) return r;

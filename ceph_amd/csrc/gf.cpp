@@ -108,8 +108,117 @@ bool gen_matrix(int technique, std::vector<uint8_t> &a, int k, int m) {
     case 0: return gen_matrix_rs_van_isa(a, k, m);
     case 1: return gen_matrix_cauchy_isa(a, k, m);
     case 2: return gen_matrix_rs_van_jerasure(a, k, m);
+    case 3: return gen_matrix_cauchy_orig(a, k, m);
     default: return false;
   }
+}
+
+// jerasure cauchy.c cauchy_original_coding_matrix: coding (i,j) =
+// inv(i XOR (m+j)).
+bool gen_matrix_cauchy_orig(std::vector<uint8_t> &a, int k, int m) {
+  const GF8 &f = gf8();
+  if (k < 1 || m < 1 || k + m > 255) return false;
+  a.assign((size_t)(k + m) * k, 0);
+  for (int i = 0; i < k; i++) a[(size_t)k * i + i] = 1;
+  for (int i = 0; i < m; i++)
+    for (int j = 0; j < k; j++) {
+      int x = i ^ (m + j);
+      if (!x) return false;
+      a[(size_t)(k + i) * k + j] = f.inv((uint8_t)x);
+    }
+  return true;
+}
+
+// jerasure.c jerasure_matrix_to_bitmatrix: block (i,j) column c = bits of
+// coeff * 2^c (companion basis).
+void matrix_to_bitmatrix(const uint8_t *coding_rows, int k, int m, int w,
+                         std::vector<uint8_t> &bitmat) {
+  const GF8 &f = gf8();
+  int W = k * w;
+  bitmat.assign((size_t)m * w * W, 0);
+  for (int i = 0; i < m; i++)
+    for (int j = 0; j < k; j++) {
+      uint8_t v = coding_rows[(size_t)i * k + j];
+      for (int c = 0; c < w; c++) {
+        for (int r = 0; r < w; r++)
+          bitmat[(size_t)(i * w + r) * W + j * w + c] = (v >> r) & 1;
+        v = f.mul(v, 2);
+      }
+    }
+}
+
+static bool gf2_invert(std::vector<uint8_t> &a, std::vector<uint8_t> &inv,
+                       int n) {
+  inv.assign((size_t)n * n, 0);
+  for (int i = 0; i < n; i++) inv[(size_t)i * n + i] = 1;
+  for (int i = 0; i < n; i++) {
+    if (!a[(size_t)i * n + i]) {
+      int j = i + 1;
+      while (j < n && !a[(size_t)j * n + i]) j++;
+      if (j >= n) return false;
+      for (int c = 0; c < n; c++) {
+        std::swap(a[(size_t)i * n + c], a[(size_t)j * n + c]);
+        std::swap(inv[(size_t)i * n + c], inv[(size_t)j * n + c]);
+      }
+    }
+    for (int r = 0; r < n; r++) {
+      if (r == i || !a[(size_t)r * n + i]) continue;
+      for (int c = 0; c < n; c++) {
+        a[(size_t)r * n + c] ^= a[(size_t)i * n + c];
+        inv[(size_t)r * n + c] ^= inv[(size_t)i * n + c];
+      }
+    }
+  }
+  return true;
+}
+
+bool compose_bit_decode_rows(const std::vector<uint8_t> &bitmat, int k,
+                             int m, int w, uint64_t present_mask,
+                             std::vector<int> &survivors,
+                             std::vector<int> &erased,
+                             std::vector<uint8_t> &rows) {
+  int n = k + m, W = k * w;
+  survivors.clear();
+  erased.clear();
+  for (int i = 0; i < n; i++) {
+    if (present_mask & (1ull << i)) {
+      if ((int)survivors.size() < k) survivors.push_back(i);
+    } else {
+      erased.push_back(i);
+    }
+  }
+  if ((int)survivors.size() < k || (int)erased.size() > m) return false;
+
+  std::vector<uint8_t> B((size_t)W * W, 0), D;
+  for (int i = 0; i < k; i++) {
+    int id = survivors[i];
+    if (id < k) {
+      for (int r = 0; r < w; r++) B[(size_t)(i * w + r) * W + id * w + r] = 1;
+    } else {
+      std::memcpy(&B[(size_t)(i * w) * W],
+                  &bitmat[(size_t)((id - k) * w) * W], (size_t)w * W);
+    }
+  }
+  if (!gf2_invert(B, D, W)) return false;
+
+  rows.assign(erased.size() * (size_t)w * W, 0);
+  for (size_t p = 0; p < erased.size(); p++) {
+    int e = erased[p];
+    uint8_t *out = &rows[p * (size_t)w * W];
+    if (e < k) {
+      std::memcpy(out, &D[(size_t)(e * w) * W], (size_t)w * W);
+    } else {
+      const uint8_t *C = &bitmat[(size_t)((e - k) * w) * W];
+      for (int r = 0; r < w; r++)
+        for (int c = 0; c < W; c++) {
+          uint8_t s = 0;
+          for (int t = 0; t < W; t++)
+            s ^= C[(size_t)r * W + t] & D[(size_t)t * W + c];
+          out[(size_t)r * W + c] = s;
+        }
+    }
+  }
+  return true;
 }
 
 bool gf_invert(const uint8_t *in, uint8_t *out, int k) {

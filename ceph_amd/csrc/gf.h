@@ -46,6 +46,22 @@ bool gen_matrix(int technique, std::vector<uint8_t> &a, int k, int m);
 // Gauss-Jordan inverse of a k x k matrix; false if singular.
 bool gf_invert(const uint8_t *in, uint8_t *out, int k);
 
+// jerasure Cauchy-original family (w=8): full generator (identity top,
+// coding row i = inv(i XOR (m+j))), companion-basis bitmatrix, GF(2)
+// survivor inversion and decode-row composition over bit rows. Mirrors
+// jerasure cauchy.c / jerasure.c semantics as called from
+// ErasureCodeJerasure.cc:499-514,568-574.
+bool gen_matrix_cauchy_orig(std::vector<uint8_t> &a, int k, int m);
+// bitmat: (m*w) x (k*w) bits, one byte per bit, row-major
+void matrix_to_bitmatrix(const uint8_t *coding_rows, int k, int m, int w,
+                         std::vector<uint8_t> &bitmat);
+// rows out: (n_erased*w) x (k*w) bit rows over the k survivors
+bool compose_bit_decode_rows(const std::vector<uint8_t> &bitmat, int k,
+                             int m, int w, uint64_t present_mask,
+                             std::vector<int> &survivors,
+                             std::vector<int> &erased,
+                             std::vector<uint8_t> &rows);
+
 // Decode-row composition for a given erasure pattern (mirrors
 // ErasureCodeIsa.cc:510-567): picks the first k present chunks in id order
 // as survivors (ErasureCode.cc:154-170), inverts the survivor submatrix,

@@ -22,13 +22,17 @@ int technique_id(const std::string &t) {
   if (t == "reed_sol_van") return ECX_T_RS_VAN_ISA;
   if (t == "cauchy") return ECX_T_CAUCHY_ISA;
   if (t == "jerasure_reed_sol_van") return ECX_T_RS_VAN_JERASURE;
+  if (t == "cauchy_orig") return ECX_T_CAUCHY_ORIG_JERASURE;
   return -1;
 }
 
 class ErasureCodeMi355x final : public ErasureCode {
   ecx_ctx *ctx_ = nullptr;
-  int k_ = 0, m_ = 0, w_ = 8, device_ = 0, streams_ = 2;
+  int k_ = 0, m_ = 0, w_ = 8, device_ = 0, streams_ = 2, packetsize_ = 2048;
   std::string technique_;
+  bool is_bitmatrix() const {
+    return technique_id(technique_) == ECX_T_CAUCHY_ORIG_JERASURE;
+  }
 
  public:
   explicit ErasureCodeMi355x(std::string technique)
@@ -43,13 +47,16 @@ class ErasureCodeMi355x final : public ErasureCode {
 
   plugin_flags get_supported_optimizations() const override {
     // the subset the conformance tests verify (cf. ErasureCodeIsa.h:68-79;
-    // CRC composition is not implemented yet, so not claimed)
-    return FLAG_EC_PLUGIN_PARTIAL_READ_OPTIMIZATION |
-           FLAG_EC_PLUGIN_PARTIAL_WRITE_OPTIMIZATION |
-           FLAG_EC_PLUGIN_ZERO_INPUT_ZERO_OUTPUT_OPTIMIZATION |
-           FLAG_EC_PLUGIN_PARITY_DELTA_OPTIMIZATION |
-           FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED |
-           FLAG_EC_PLUGIN_DIRECT_READS;
+    // CRC composition is not implemented yet, so not claimed); parity-delta
+    // is matrix-technique-only (cauchy_orig delta schedules are a later
+    // round — claim only what works, the conformance rule)
+    plugin_flags f = FLAG_EC_PLUGIN_PARTIAL_READ_OPTIMIZATION |
+                     FLAG_EC_PLUGIN_PARTIAL_WRITE_OPTIMIZATION |
+                     FLAG_EC_PLUGIN_ZERO_INPUT_ZERO_OUTPUT_OPTIMIZATION |
+                     FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED |
+                     FLAG_EC_PLUGIN_DIRECT_READS;
+    if (!is_bitmatrix()) f |= FLAG_EC_PLUGIN_PARITY_DELTA_OPTIMIZATION;
+    return f;
   }
 
   int parse(ErasureCodeProfile &profile, std::ostream *ss) {
@@ -59,6 +66,8 @@ class ErasureCodeMi355x final : public ErasureCode {
     err |= to_int("w", profile, &w_, "8", ss);
     err |= to_int("mi355x-device", profile, &device_, "0", ss);
     err |= to_int("mi355x-streams", profile, &streams_, "2", ss);
+    if (is_bitmatrix())
+      err |= to_int("packetsize", profile, &packetsize_, "2048", ss);
     err |= sanity_check_k_m(k_, m_, ss);
     if (w_ != 8) {
       if (ss) *ss << "mi355x: w=" << w_ << " must be 8\n";
@@ -75,8 +84,8 @@ class ErasureCodeMi355x final : public ErasureCode {
   int init(ErasureCodeProfile &profile, std::ostream *ss) override {
     int err = parse(profile, ss);
     if (err) return err;
-    int r = ecx_create(k_, m_, technique_id(technique_), device_, streams_,
-                       &ctx_);
+    int r = ecx_create2(k_, m_, technique_id(technique_), w_, packetsize_,
+                        device_, streams_, &ctx_);
     if (r != ECX_OK) {
       if (ss)
         *ss << "mi355x: ecx_create failed (" << r
@@ -163,6 +172,9 @@ class ErasureCodeMi355x final : public ErasureCode {
 
   void apply_delta(const shard_id_map<buffer> &in,
                    shard_id_map<buffer> &out) override {
+    if (is_bitmatrix())
+      throw std::runtime_error(
+          "apply_delta not supported for cauchy_orig (flag not claimed)");
     // loop structure mirrors isa apply_delta (ErasureCodeIsa.cc:333-366)
     for (auto &&[datashard, databuf] : in) {
       if ((int)datashard >= k_) continue;

@@ -23,13 +23,16 @@ int technique_id(const std::string &t) {
   if (t == "reed_sol_van") return ECREF_T_RS_VAN_ISA;
   if (t == "cauchy") return ECREF_T_CAUCHY_ISA;
   if (t == "jerasure_reed_sol_van") return ECREF_T_RS_VAN_JERASURE;
+  if (t == "cauchy_orig") return 3;  // bitmatrix/packet layout
   return -1;
 }
 
 class ErasureCodeOracle final : public ErasureCode {
-  int k_ = 0, m_ = 0;
+  int k_ = 0, m_ = 0, packetsize_ = 2048, w_ = 8;
   std::string technique_;
   std::vector<uint8_t> gen_;
+  std::vector<uint8_t> bitmat_;
+  bool is_bitmatrix() const { return technique_ == "cauchy_orig"; }
 
  public:
   explicit ErasureCodeOracle(std::string t) : technique_(std::move(t)) {}
@@ -55,14 +58,36 @@ class ErasureCodeOracle final : public ErasureCode {
       if (ss) *ss << "oracle: unknown technique " << technique_ << "\n";
       err = -EINVAL;
     }
+    if (is_bitmatrix())
+      err |= to_int("packetsize", profile, &packetsize_, "2048", ss);
     if (err) return err;
     profile["technique"] = technique_;
     gen_.resize((size_t)(k_ + m_) * k_);
-    if (ecref_matrix(t, gen_.data(), k_, m_) != 0) return -EINVAL;
+    if (is_bitmatrix()) {
+      std::vector<uint8_t> coding((size_t)m_ * k_);
+      if (ecref_matrix_cauchy_orig_jerasure(coding.data(), k_, m_) != 0)
+        return -EINVAL;
+      for (int i = 0; i < k_; i++) {
+        std::memset(&gen_[(size_t)i * k_], 0, k_);
+        gen_[(size_t)i * k_ + i] = 1;
+      }
+      std::memcpy(&gen_[(size_t)k_ * k_], coding.data(), (size_t)m_ * k_);
+      bitmat_.resize((size_t)m_ * w_ * k_ * w_);
+      ecref_matrix_to_bitmatrix(coding.data(), k_, m_, w_, bitmat_.data());
+    } else if (ecref_matrix(t, gen_.data(), k_, m_) != 0) {
+      return -EINVAL;
+    }
     return ErasureCode::init(profile, ss);
   }
 
   unsigned int get_chunk_size(unsigned int stripe_width) const override {
+    if (is_bitmatrix()) {
+      // ErasureCodeJerasureCauchy::get_alignment rule
+      unsigned align = (unsigned)k_ * w_ * packetsize_ * 4u;
+      unsigned tail = stripe_width % align;
+      unsigned padded = stripe_width + (tail ? align - tail : 0);
+      return padded / k_;
+    }
     if (technique_ == "jerasure_reed_sol_van")
       return ecref_chunk_size_jerasure(k_, 8, stripe_width);
     return ecref_chunk_size_isa(k_, stripe_width);
@@ -96,6 +121,9 @@ class ErasureCodeOracle final : public ErasureCode {
       }
     }
     if (!size) return 0;
+    if (is_bitmatrix())
+      return ecref_bitmatrix_encode(k_, m_, w_, bitmat_.data(), data, parity,
+                                    size, packetsize_) == 0 ? 0 : -EINVAL;
     ecref_encode(k_, m_, gen_.data() + (size_t)k_ * k_, data, parity, size);
     return 0;
   }
@@ -125,6 +153,11 @@ class ErasureCodeOracle final : public ErasureCode {
         chunks[i] = scratch.back().c_str();
       }
     }
+    if (is_bitmatrix())
+      return ecref_bitmatrix_decode(k_, m_, w_, bitmat_.data(), chunks,
+                                    present, size, packetsize_) == 0
+                 ? 0
+                 : -EIO;
     return ecref_decode(technique_id(technique_), k_, m_, chunks, present,
                         size) == 0
                ? 0

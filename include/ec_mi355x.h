@@ -59,6 +59,11 @@ enum ecx_technique {
   ECX_T_RS_VAN_ISA = 0,
   ECX_T_CAUCHY_ISA = 1,
   ECX_T_RS_VAN_JERASURE = 2,
+  /* jerasure cauchy_orig: bitmatrix/packet layout (w=8), profile key
+   * packetsize (default 2048, ErasureCodeJerasure.h DEFAULT_PACKETSIZE);
+   * chunk sizes must be multiples of w*packetsize
+   * (ErasureCodeJerasureCauchy::get_alignment, ErasureCodeJerasure.cc:522-536) */
+  ECX_T_CAUCHY_ORIG_JERASURE = 3,
 };
 
 enum ecx_err {
@@ -86,6 +91,10 @@ ECX_API int ecx_device_count(void);
  * mirroring prepare() (ErasureCodeIsa.cc:637-697). */
 ECX_API int ecx_create(int k, int m, int technique, int device, int n_streams,
                        ecx_ctx **out);
+/* Extended form: w (must be 8) and packetsize (bitmatrix techniques only;
+ * ignored for matrix techniques). */
+ECX_API int ecx_create2(int k, int m, int technique, int w, int packetsize,
+                        int device, int n_streams, ecx_ctx **out);
 ECX_API void ecx_destroy(ecx_ctx *ctx);
 
 ECX_API int ecx_k(const ecx_ctx *ctx);

@@ -16,10 +16,12 @@ _DIR = os.path.dirname(os.path.abspath(__file__))
 T_RS_VAN_ISA = 0
 T_CAUCHY_ISA = 1
 T_RS_VAN_JERASURE = 2
+T_CAUCHY_ORIG_JERASURE = 3  # bitmatrix/packet layout: use bitmatrix_* fns
 TECHNIQUES = {
     "reed_sol_van": T_RS_VAN_ISA,
     "cauchy": T_CAUCHY_ISA,
     "jerasure_reed_sol_van": T_RS_VAN_JERASURE,
+    "cauchy_orig": T_CAUCHY_ORIG_JERASURE,
 }
 
 
@@ -59,6 +61,9 @@ _ref.ecref_chunk_size_isa.argtypes = [ctypes.c_int, ctypes.c_uint]
 _ref.ecref_chunk_size_jerasure.restype = ctypes.c_uint
 _ref.ecref_chunk_size_jerasure.argtypes = [ctypes.c_int, ctypes.c_int,
                                            ctypes.c_uint]
+_ref.ecref_matrix_cauchy_orig_jerasure.restype = ctypes.c_int
+_ref.ecref_bitmatrix_encode.restype = ctypes.c_int
+_ref.ecref_bitmatrix_decode.restype = ctypes.c_int
 _cpu.eccpu_encode_batch.restype = ctypes.c_int
 _cpu.eccpu_decode_batch.restype = ctypes.c_int
 _cpu.eccpu_threads.restype = ctypes.c_int
@@ -83,6 +88,11 @@ def gf_exp_table():
 def matrix(technique, k, m):
     """Full (k+m) x k generator (identity top), as uint8 ndarray."""
     t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    if t == T_CAUCHY_ORIG_JERASURE:
+        a = np.zeros((k + m, k), dtype=np.uint8)
+        a[:k] = np.eye(k, dtype=np.uint8)
+        a[k:] = cauchy_orig_matrix(k, m)
+        return a
     a = np.zeros((k + m, k), dtype=np.uint8)
     r = _ref.ecref_matrix(t, a.ctypes.data_as(ctypes.c_void_p), k, m)
     if r != 0:
@@ -100,6 +110,10 @@ def _ptr_array(bufs):
 def encode(technique, k, m, data, chunk_bytes=None):
     """data: list of k uint8 arrays (or None for zeros). Returns list of m
     parity arrays. Scalar oracle (ecref_encode)."""
+    t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    if t == T_CAUCHY_ORIG_JERASURE:
+        raise ValueError("cauchy_orig uses the bitmatrix/packet layout: "
+                         "call bitmatrix_encode/bitmatrix_decode")
     lens = {d.nbytes for d in data if d is not None}
     assert len(lens) == 1 or (not lens and chunk_bytes)
     length = lens.pop() if lens else chunk_bytes
@@ -123,6 +137,53 @@ def decode(technique, k, m, chunks, present):
                           ctypes.c_size_t(length))
     if r != 0:
         raise ValueError(f"ecref_decode failed: {r}")
+    return chunks
+
+
+def cauchy_orig_matrix(k, m):
+    """jerasure cauchy_orig coding matrix (m x k): inv(i XOR (m+j))."""
+    a = np.zeros((m, k), dtype=np.uint8)
+    r = _ref.ecref_matrix_cauchy_orig_jerasure(
+        a.ctypes.data_as(ctypes.c_void_p), k, m)
+    if r != 0:
+        raise ValueError(f"cauchy_orig matrix failed: {r}")
+    return a
+
+
+def bitmatrix(coding, w=8):
+    m, k = coding.shape
+    bm = np.zeros((m * w, k * w), dtype=np.uint8)
+    _ref.ecref_matrix_to_bitmatrix(
+        np.ascontiguousarray(coding).ctypes.data_as(ctypes.c_void_p), k, m,
+        w, bm.ctypes.data_as(ctypes.c_void_p))
+    return bm
+
+
+def bitmatrix_encode(k, m, data, packetsize, w=8):
+    """jerasure cauchy_orig (bitmatrix/packet layout) encode; data entries
+    may be None for zeros chunks."""
+    lens = {d.nbytes for d in data if d is not None}
+    assert len(lens) == 1
+    size = lens.pop()
+    bm = bitmatrix(cauchy_orig_matrix(k, m), w)
+    parity = [np.zeros(size, dtype=np.uint8) for _ in range(m)]
+    r = _ref.ecref_bitmatrix_encode(
+        k, m, w, bm.ctypes.data_as(ctypes.c_void_p), _ptr_array(data),
+        _ptr_array(parity), ctypes.c_size_t(size), packetsize)
+    if r != 0:
+        raise ValueError(f"bitmatrix_encode failed: {r}")
+    return parity
+
+
+def bitmatrix_decode(k, m, chunks, present, packetsize, w=8):
+    bm = bitmatrix(cauchy_orig_matrix(k, m), w)
+    pres = np.asarray(present, dtype=np.uint8)
+    r = _ref.ecref_bitmatrix_decode(
+        k, m, w, bm.ctypes.data_as(ctypes.c_void_p), _ptr_array(chunks),
+        pres.ctypes.data_as(ctypes.c_void_p),
+        ctypes.c_size_t(chunks[0].nbytes), packetsize)
+    if r != 0:
+        raise ValueError(f"bitmatrix_decode failed: {r}")
     return chunks
 
 

@@ -3,6 +3,7 @@
 #include "ec_ref.h"
 
 #include <errno.h>
+#include <stdlib.h>
 #include <string.h>
 
 /* ---------------- GF(2^8) over 0x11d ---------------- */
@@ -396,4 +397,191 @@ unsigned ecref_chunk_size_jerasure(int k, int w, unsigned stripe_width)
   unsigned tail = stripe_width % alignment;
   unsigned padded = stripe_width + (tail ? alignment - tail : 0);
   return padded / k;
+}
+
+/* ---------------- jerasure bitmatrix (Cauchy-original) family ----------- */
+
+/* cauchy.c cauchy_original_coding_matrix: m[i][j] = 1/(i XOR (m+j)). */
+int ecref_matrix_cauchy_orig_jerasure(uint8_t *coding, int k, int m)
+{
+  ecref_gf_init();
+  if (k < 1 || m < 1 || k + m > 255)
+    return -EINVAL;
+  for (int i = 0; i < m; i++)
+    for (int j = 0; j < k; j++) {
+      int x = i ^ (m + j);
+      if (x == 0)
+        return -EDOM;
+      coding[(size_t)i * k + j] = ecref_gf_inv((uint8_t)x);
+    }
+  return 0;
+}
+
+/* jerasure.c jerasure_matrix_to_bitmatrix: block (i,j) column c holds the
+ * bit pattern of coeff * 2^c (companion-matrix representation): bit row r
+ * of the block = bit r of gf_mul(coeff, 1<<c). */
+void ecref_matrix_to_bitmatrix(const uint8_t *coding, int k, int m, int w,
+                               uint8_t *bitmat)
+{
+  ecref_gf_init();
+  int W = k * w;
+  for (int i = 0; i < m; i++)
+    for (int j = 0; j < k; j++) {
+      uint8_t v = coding[(size_t)i * k + j];
+      for (int c = 0; c < w; c++) {
+        for (int r = 0; r < w; r++)
+          bitmat[(size_t)(i * w + r) * W + j * w + c] = (v >> r) & 1;
+        v = ecref_gf_mul(v, 2);
+      }
+    }
+}
+
+static void bitmatrix_dotprod(int n_src, int w, const uint8_t *rows /* w x n_src*w */,
+                              const uint8_t *const *src, uint8_t *dst,
+                              size_t size, int packetsize)
+{
+  size_t super = (size_t)w * packetsize;
+  for (size_t off = 0; off < size; off += super) {
+    for (int r = 0; r < w; r++) {
+      uint8_t *d = dst + off + (size_t)r * packetsize;
+      memset(d, 0, packetsize);
+      const uint8_t *row = rows + (size_t)r * n_src * w;
+      for (int j = 0; j < n_src; j++) {
+        if (src[j] == NULL)
+          continue; /* zeros chunk */
+        for (int c = 0; c < w; c++) {
+          if (!row[j * w + c])
+            continue;
+          const uint8_t *s = src[j] + off + (size_t)c * packetsize;
+          for (int b = 0; b < packetsize; b++)
+            d[b] ^= s[b];
+        }
+      }
+    }
+  }
+}
+
+int ecref_bitmatrix_encode(int k, int m, int w, const uint8_t *bitmat,
+                           const uint8_t *const *data, uint8_t *const *coding,
+                           size_t size, int packetsize)
+{
+  if (packetsize <= 0 || size % ((size_t)w * packetsize))
+    return -EINVAL;
+  for (int i = 0; i < m; i++)
+    bitmatrix_dotprod(k, w, bitmat + (size_t)(i * w) * k * w, data,
+                      coding[i], size, packetsize);
+  return 0;
+}
+
+/* GF(2) Gauss-Jordan inversion of an n x n bit matrix (bytes 0/1). */
+static int gf2_invert(uint8_t *a, uint8_t *inv, int n)
+{
+  memset(inv, 0, (size_t)n * n);
+  for (int i = 0; i < n; i++)
+    inv[(size_t)i * n + i] = 1;
+  for (int i = 0; i < n; i++) {
+    if (!a[(size_t)i * n + i]) {
+      int j = i + 1;
+      while (j < n && !a[(size_t)j * n + i])
+        j++;
+      if (j >= n)
+        return -1;
+      for (int c = 0; c < n; c++) {
+        uint8_t t = a[(size_t)i * n + c];
+        a[(size_t)i * n + c] = a[(size_t)j * n + c];
+        a[(size_t)j * n + c] = t;
+        t = inv[(size_t)i * n + c];
+        inv[(size_t)i * n + c] = inv[(size_t)j * n + c];
+        inv[(size_t)j * n + c] = t;
+      }
+    }
+    for (int r = 0; r < n; r++) {
+      if (r == i || !a[(size_t)r * n + i])
+        continue;
+      for (int c = 0; c < n; c++) {
+        a[(size_t)r * n + c] ^= a[(size_t)i * n + c];
+        inv[(size_t)r * n + c] ^= inv[(size_t)i * n + c];
+      }
+    }
+  }
+  return 0;
+}
+
+int ecref_bitmatrix_decode(int k, int m, int w, const uint8_t *bitmat,
+                           uint8_t *const *chunks, const uint8_t *present,
+                           size_t size, int packetsize)
+{
+  if (packetsize <= 0 || size % ((size_t)w * packetsize))
+    return -EINVAL;
+  int n = k + m, W = k * w;
+  int nerrs = 0, erasures[255], decode_index[255];
+  for (int i = 0; i < n; i++)
+    if (!present[i])
+      erasures[nerrs++] = i;
+  if (nerrs == 0)
+    return 0;
+  if (nerrs > m)
+    return -1;
+  {
+    int r = 0;
+    for (int i = 0; i < k; i++, r++) {
+      while (r < n && !present[r])
+        r++;
+      if (r >= n)
+        return -1;
+      decode_index[i] = r;
+    }
+  }
+  /* survivor bit matrix B (W x W): data survivor -> identity block row,
+   * coding survivor -> its bitmatrix rows */
+  /* casts keep this file compilable as C and as C++ (the oracle fixture
+   * plugin builds it with g++) */
+  uint8_t *B = (uint8_t *)malloc((size_t)W * W);
+  uint8_t *D = (uint8_t *)malloc((size_t)W * W);
+  uint8_t *rows = (uint8_t *)malloc((size_t)m * w * W);
+  if (!B || !D || !rows) {
+    free(B); free(D); free(rows);
+    return -ENOMEM;
+  }
+  memset(B, 0, (size_t)W * W);
+  for (int i = 0; i < k; i++) {
+    int id = decode_index[i];
+    if (id < k) {
+      for (int r = 0; r < w; r++)
+        B[(size_t)(i * w + r) * W + id * w + r] = 1;
+    } else {
+      memcpy(&B[(size_t)(i * w) * W], &bitmat[(size_t)((id - k) * w) * W],
+             (size_t)w * W);
+    }
+  }
+  if (gf2_invert(B, D, W) != 0) {
+    free(B); free(D); free(rows);
+    return -1;
+  }
+  const uint8_t *src[255];
+  for (int i = 0; i < k; i++)
+    src[i] = chunks[decode_index[i]];
+
+  for (int p = 0; p < nerrs; p++) {
+    int e = erasures[p];
+    uint8_t *out_rows = rows + (size_t)p * w * W;
+    if (e < k) {
+      memcpy(out_rows, &D[(size_t)(e * w) * W], (size_t)w * W);
+    } else {
+      /* lost coding: compose its bitmatrix rows with D over GF(2) */
+      const uint8_t *C = &bitmat[(size_t)((e - k) * w) * W];
+      for (int r = 0; r < w; r++)
+        for (int c = 0; c < W; c++) {
+          uint8_t s = 0;
+          for (int t = 0; t < W; t++)
+            s ^= C[(size_t)r * W + t] & D[(size_t)t * W + c];
+          out_rows[(size_t)r * W + c] = s;
+        }
+    }
+  }
+  for (int p = 0; p < nerrs; p++)
+    bitmatrix_dotprod(k, w, rows + (size_t)p * w * W, src,
+                      chunks[erasures[p]], size, packetsize);
+  free(B); free(D); free(rows);
+  return 0;
 }

@@ -113,6 +113,34 @@ void ecref_region_mul_xor(uint8_t coeff, const uint8_t *delta, uint8_t *parity,
 unsigned ecref_chunk_size_isa(int k, unsigned stripe_width);       /* align 32 */
 unsigned ecref_chunk_size_jerasure(int k, int w, unsigned stripe_width);
 
+/* ---- jerasure bitmatrix (Cauchy-original) family, w=8 ----
+ * Restates jerasure cauchy.c cauchy_original_coding_matrix (m[i][j] =
+ * 1/(i XOR (m+j))), jerasure.c jerasure_matrix_to_bitmatrix (w x w
+ * companion-matrix block per coefficient: block column c holds the bits of
+ * coeff*2^c), and the bitmatrix encode/decode data layout used by
+ * jerasure_schedule_encode / jerasure_schedule_decode_lazy as called from
+ * ErasureCodeJerasure.cc:499-514: each chunk is processed in superwords of
+ * w*packetsize bytes; packet row r of a coding superword is the XOR of the
+ * data packets selected by bitmatrix row r. (Smart/dumb schedules are just
+ * XOR orderings of the same linear map; the bytes are identical.)
+ * Same parity-pinning caveats as the header note: the jerasure sources are
+ * absent; pinned by structure (companion-basis GF equivalence, tested) and
+ * round-trip/property tests.
+ */
+int  ecref_matrix_cauchy_orig_jerasure(uint8_t *coding /* m x k */, int k, int m);
+/* bitmat: (m*w) x (k*w) entries, one byte per bit, row-major */
+void ecref_matrix_to_bitmatrix(const uint8_t *coding, int k, int m, int w,
+                               uint8_t *bitmat);
+/* size must be a multiple of w*packetsize; data[i] may be NULL (zeros) */
+int  ecref_bitmatrix_encode(int k, int m, int w, const uint8_t *bitmat,
+                            const uint8_t *const *data, uint8_t *const *coding,
+                            size_t size, int packetsize);
+/* decode under erasures; chunks[] has k+m entries, erased ones are
+ * reconstructed in place (survivors = first k present in id order). */
+int  ecref_bitmatrix_decode(int k, int m, int w, const uint8_t *bitmat,
+                            uint8_t *const *chunks, const uint8_t *present,
+                            size_t size, int packetsize);
+
 #ifdef __cplusplus
 }
 #endif

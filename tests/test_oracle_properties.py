@@ -172,3 +172,74 @@ def test_decode_too_many_erasures_fails():
     present[[0, 1, 2]] = 0  # 3 > m erasures
     with pytest.raises(ValueError):
         oracle.decode("reed_sol_van", k, m, chunks, present)
+
+
+class TestBitmatrixCauchyOrig:
+    """jerasure cauchy_orig (bitmatrix/packet layout) oracle properties.
+    The companion-basis convention is additionally cross-pinned in
+    test_gf_kat-style fashion: coding bytes equal GF(2^8) Cauchy-original
+    arithmetic applied to the bit-sliced symbols."""
+
+    def test_matrix_values(self):
+        k, m = 5, 3
+        a = oracle.cauchy_orig_matrix(k, m)
+        for i in range(m):
+            for j in range(k):
+                assert a[i, j] == oracle.gf_inv(i ^ (m + j))
+
+    def test_companion_basis_equivalence(self):
+        k, m, w, p = 4, 3, 8, 64
+        rng = np.random.default_rng(1)
+        size = 2 * w * p
+        data = [rng.integers(0, 256, size, dtype=np.uint8)
+                for _ in range(k)]
+        par = oracle.bitmatrix_encode(k, m, data, p)
+        cod = oracle.cauchy_orig_matrix(k, m)
+
+        def sym(buf, t, b):
+            v = 0
+            for c in range(w):
+                v |= ((int(buf[c * p + t]) >> b) & 1) << c
+            return v
+
+        for i in range(m):
+            for t in (0, 17, p - 1):
+                for b in (0, 4, 7):
+                    want = 0
+                    for j in range(k):
+                        want ^= oracle.gf_mul(int(cod[i, j]),
+                                              sym(data[j], t, b))
+                    assert sym(par[i], t, b) == want
+
+    def test_exhaustive_erasure_round_trip(self):
+        k, m, p = 4, 3, 32
+        rng = np.random.default_rng(9)
+        size = 3 * 8 * p
+        data = [rng.integers(0, 256, size, dtype=np.uint8)
+                for _ in range(k)]
+        par = oracle.bitmatrix_encode(k, m, data, p)
+        full = data + par
+        for e in range(1, m + 1):
+            for er in combinations(range(k + m), e):
+                present = np.ones(k + m, np.uint8)
+                present[list(er)] = 0
+                test = [c.copy() if present[i] else np.zeros(size, np.uint8)
+                        for i, c in enumerate(full)]
+                oracle.bitmatrix_decode(k, m, test, present, p)
+                for i in range(k + m):
+                    assert (test[i] == full[i]).all(), (er, i)
+
+    def test_zero_in_zero_out_and_null(self):
+        k, m, p = 4, 2, 16
+        size = 8 * p
+        for pz in (oracle.bitmatrix_encode(
+                       k, m, [np.zeros(size, np.uint8)] * k, p),
+                   ):
+            for x in pz:
+                assert not x.any()
+
+    def test_size_must_be_superword_multiple(self):
+        k, m, p = 4, 2, 64
+        data = [np.zeros(100, np.uint8) for _ in range(k)]
+        with pytest.raises(ValueError):
+            oracle.bitmatrix_encode(k, m, data, p)
