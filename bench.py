@@ -68,7 +68,10 @@ def parity_selfcheck(ctx, dptr, args, seed, sample_stripes=2):
         exp = expected_fill(int(s) * stripe_bytes, k * C, seed)
         if not np.array_equal(host[:k * C], exp):
             raise AssertionError(f"stripe {s}: data region != expected fill")
-        want = oracle.encode(args.technique, k, m, data)
+        if args.technique == "cauchy_orig":
+            want = oracle.bitmatrix_encode(k, m, data, 2048)
+        else:
+            want = oracle.encode(args.technique, k, m, data)
         for j in range(m):
             got = host[(k + j) * C:(k + j + 1) * C]
             if not np.array_equal(got, want[j]):
@@ -382,7 +385,8 @@ def main():
                 "alg_bytes_per_launch": alg_bytes,
             },
             "cpu_baseline": (cpu_baseline(args)
-                             if (world == 1 and not args.no_cpu_baseline)
+                             if (world == 1 and not args.no_cpu_baseline
+                                 and args.technique != "cauchy_orig")
                              else None),
         }
         del line["config"]["encode_gibs"]

@@ -1,4 +1,5 @@
 // plugin_lrc.cc — libec_lrc.so: locally-repairable codes as layered
+#include <algorithm>
 // composition over registry sub-plugins, mirroring the reference's LRC
 // plugin (src/erasure-code/lrc/ErasureCodeLrc.cc): parse_kml expansion
 // (:292-395), layer chunk-map parsing and sub-plugin instantiation via the
