@@ -1,5 +1,4 @@
 // plugin_lrc.cc — libec_lrc.so: locally-repairable codes as layered
-#include <algorithm>
 // composition over registry sub-plugins, mirroring the reference's LRC
 // plugin (src/erasure-code/lrc/ErasureCodeLrc.cc): parse_kml expansion
 // (:292-395), layer chunk-map parsing and sub-plugin instantiation via the
@@ -13,6 +12,7 @@
 // parse_kml REJECTS it ((k+m) % l != 0, ERROR_LRC_K_M_MODULO); the bench
 // uses the nearest valid shape (k=9 m=3 l=4 or the doc example k=4 m=2
 // l=3) and says so.
+#include <algorithm>
 #include <cerrno>
 #include <cstring>
 #include <memory>
