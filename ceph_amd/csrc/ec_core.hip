@@ -333,24 +333,36 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   }
   __syncthreads();
 
-  for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
-    const int r = t >> vq_shift;
-    const int v = t - (r << vq_shift);
-    v4u acc = {0, 0, 0, 0};
+  // One row per wave: the row's op list is cached across the wave's lanes
+  // (lane l holds ops b0+l, b0+64+l) and walked with v_readlane — no
+  // per-term LDS op reads, and the row branch stays wave-uniform.
+  (void)vq_shift;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int waves = blockDim.x >> 6;
+  for (int r = wave; r < n_rows; r += waves) {
     const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
-    for (int o = b0; o < b1; o++) {
-      const int jc = s_ops[o];
-      const v4u d =
-          *reinterpret_cast<const v4u*>(s_data + (size_t)jc * q + (size_t)v * 16);
-      acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
+    const int n_row_ops = b1 - b0;  // <= k*w <= 128
+    uint32_t op0 = (b0 + lane < b1) ? (uint32_t)s_ops[b0 + lane] : 0u;
+    uint32_t op1 =
+        (b0 + 64 + lane < b1) ? (uint32_t)s_ops[b0 + 64 + lane] : 0u;
+    uint8_t* drow = obase + (long)bp->out_ids[r / w] * chunk_bytes +
+                    sw * (long)w * pkt + (long)(r % w) * pkt + (long)win * q;
+    for (int v = lane; v < vq; v += 64) {
+      v4u acc = {0, 0, 0, 0};
+      for (int o = 0; o < n_row_ops; o++) {
+        const int jc = (o < 64) ? __builtin_amdgcn_readlane(op0, o)
+                                : __builtin_amdgcn_readlane(op1, o - 64);
+        const v4u d = *reinterpret_cast<const v4u*>(
+            s_data + (size_t)jc * q + (size_t)v * 16);
+        acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
+      }
+      v4u* dst = reinterpret_cast<v4u*>(drow + (long)v * 16);
+      if (NT)
+        __builtin_nontemporal_store(acc, dst);
+      else
+        *dst = acc;
     }
-    v4u* dst = reinterpret_cast<v4u*>(
-        obase + (long)bp->out_ids[r / w] * chunk_bytes + sw * (long)w * pkt +
-        (long)(r % w) * pkt + (long)win * q + (long)v * 16);
-    if (NT)
-      __builtin_nontemporal_store(acc, dst);
-    else
-      *dst = acc;
   }
 }
 
@@ -931,10 +943,19 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
     return v ? atoi(v) : 1;
   }();
 
-  // LDS window: largest power-of-two divisor of pkt with n_src*w*q <= 48 KB
+  // LDS window: largest power-of-two divisor of pkt within the LDS budget
+  // (ECX_BITQ KB, default 48 — A/B-able; bigger windows fill waves better
+  // but cost residency)
+  static const size_t lds_budget = [] {
+    const char* v = getenv("ECX_BITQ");
+    long kb = v ? atol(v) : 48;
+    if (kb < 8) kb = 8;
+    if (kb > 120) kb = 120;
+    return (size_t)kb * 1024;
+  }();
   int q = 16;
   while (q * 2 <= pkt && pkt % (q * 2) == 0 &&
-         (size_t)n_src * w * q * 2 <= 48 * 1024)
+         (size_t)n_src * w * q * 2 <= lds_budget)
     q *= 2;
   int vq_shift = 0;
   while ((1 << vq_shift) < q / 16) vq_shift++;
