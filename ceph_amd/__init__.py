@@ -83,6 +83,9 @@ def _lib():
                                       ctypes.POINTER(ctypes.c_size_t),
                                       ctypes.c_int, ctypes.c_uint64,
                                       ctypes.c_int]
+    lib.ecx_set_matrix.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    lib.ecx_shec_matrix.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                                    ctypes.c_int, ctypes.c_void_p]
     lib.ecx_sync.argtypes = [ctypes.c_void_p, ctypes.c_int]
     lib.ecx_last_kernel_ms.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                        ctypes.POINTER(ctypes.c_double)]
@@ -115,6 +118,15 @@ def version():
 
 def device_count():
     return lib().ecx_device_count()
+
+
+def shec_matrix(k, m, c, single=False):
+    """SHEC shingled coding matrix (host-side; no GPU needed)."""
+    out = np.zeros((m, k), dtype=np.uint8)
+    _ck(lib().ecx_shec_matrix(k, m, c, int(single),
+                              out.ctypes.data_as(ctypes.c_void_p)),
+        "ecx_shec_matrix")
+    return out
 
 
 def _ck(r, what):
@@ -224,6 +236,14 @@ class EcContext:
         arr, sz, n = self._slice_args(chunk_ptrs, sizes)
         _ck(lib().ecx_decode_slices(self._h, arr, sz, n, present_mask,
                                     slot), "ecx_decode_slices")
+
+    def set_matrix(self, coding_rows):
+        """Replace the coding rows (custom-matrix codecs, e.g. SHEC)."""
+        rows = np.ascontiguousarray(coding_rows, dtype=np.uint8)
+        assert rows.shape == (self.m, self.k)
+        _ck(lib().ecx_set_matrix(self._h,
+                                 rows.ctypes.data_as(ctypes.c_void_p)),
+            "ecx_set_matrix")
 
     def sync(self, slot=0):
         _ck(lib().ecx_sync(self._h, slot), "ecx_sync")

@@ -333,36 +333,28 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   }
   __syncthreads();
 
-  // One row per wave: the row's op list is cached across the wave's lanes
-  // (lane l holds ops b0+l, b0+64+l) and walked with v_readlane — no
-  // per-term LDS op reads, and the row branch stays wave-uniform.
-  (void)vq_shift;
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  const int waves = blockDim.x >> 6;
-  for (int r = wave; r < n_rows; r += waves) {
+  // Item-parallel compute (one 16B vec of one output row per item): A/B
+  // showed this beats a row-per-wave readlane variant — with q=512 the
+  // row-per-wave form idles half of each wave (vq=32) and larger q
+  // collapses residency; LDS op reads broadcast cheaply.
+  for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
+    const int r = t >> vq_shift;
+    const int v = t - (r << vq_shift);
+    v4u acc = {0, 0, 0, 0};
     const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
-    const int n_row_ops = b1 - b0;  // <= k*w <= 128
-    uint32_t op0 = (b0 + lane < b1) ? (uint32_t)s_ops[b0 + lane] : 0u;
-    uint32_t op1 =
-        (b0 + 64 + lane < b1) ? (uint32_t)s_ops[b0 + 64 + lane] : 0u;
-    uint8_t* drow = obase + (long)bp->out_ids[r / w] * chunk_bytes +
-                    sw * (long)w * pkt + (long)(r % w) * pkt + (long)win * q;
-    for (int v = lane; v < vq; v += 64) {
-      v4u acc = {0, 0, 0, 0};
-      for (int o = 0; o < n_row_ops; o++) {
-        const int jc = (o < 64) ? __builtin_amdgcn_readlane(op0, o)
-                                : __builtin_amdgcn_readlane(op1, o - 64);
-        const v4u d = *reinterpret_cast<const v4u*>(
-            s_data + (size_t)jc * q + (size_t)v * 16);
-        acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
-      }
-      v4u* dst = reinterpret_cast<v4u*>(drow + (long)v * 16);
-      if (NT)
-        __builtin_nontemporal_store(acc, dst);
-      else
-        *dst = acc;
+    for (int o = b0; o < b1; o++) {
+      const int jc = s_ops[o];
+      const v4u d = *reinterpret_cast<const v4u*>(
+          s_data + (size_t)jc * q + (size_t)v * 16);
+      acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
     }
+    v4u* dst = reinterpret_cast<v4u*>(
+        obase + (long)bp->out_ids[r / w] * chunk_bytes + sw * (long)w * pkt +
+        (long)(r % w) * pkt + (long)win * q + (long)v * 16);
+    if (NT)
+      __builtin_nontemporal_store(acc, dst);
+    else
+      *dst = acc;
   }
 }
 
@@ -1158,6 +1150,74 @@ int ecx_decode_slices(ecx_ctx* ctx, void* const* d_chunks,
   return run_slices(ctx, slot, d_chunks, bytes, n_slices,
                     plan.survivors.data(), ctx->k, plan.erased.data(),
                     (int)plan.erased.size(), plan.rows.data());
+}
+
+// defined in the host-pointer section below
+static int ensure_stage(ecx_ctx* ctx, Slot& s, size_t bytes);
+
+int ecx_set_matrix(ecx_ctx* ctx, const uint8_t* coding_rows) {
+  if (!ctx || !coding_rows || ctx->is_bitmatrix()) return ECX_ERR_INVAL;
+  std::lock_guard<std::mutex> g(ctx->lru_mu);
+  std::memcpy(ctx->gen.data() + (size_t)ctx->k * ctx->k, coding_rows,
+              (size_t)ctx->m * ctx->k);
+  ctx->lru.clear();
+  ctx->lru_order.clear();
+  ctx->bit_lru.clear();
+  return ECX_OK;
+}
+
+int ecx_matmul_chunks_host(ecx_ctx* ctx, const uint8_t* const* srcs,
+                           int n_src, uint8_t* const* outs, int n_out,
+                           const uint8_t* rows, size_t bytes) {
+  if (!ctx || !srcs || !outs || !rows || n_src < 1 || n_src > ECX_MAX_K ||
+      n_out < 1 || n_out > ECX_MAX_K || bytes % 16)
+    return ECX_ERR_INVAL;
+  Slot& s = ctx->slots[0];
+  std::lock_guard<std::recursive_mutex> g(s.mu);
+  int r = ensure_stage(ctx, s, (size_t)(n_src + n_out) * bytes);
+  if (r != ECX_OK) return r;
+  HIP_TRY(hipSetDevice(ctx->device));
+  bool src_null[ECX_MAX_K] = {};
+  for (int i = 0; i < n_src; i++) {
+    if (!srcs[i]) {
+      src_null[i] = true;
+      continue;
+    }
+    HIP_TRY(hipMemcpyAsync(s.d_stage + (size_t)i * bytes, srcs[i], bytes,
+                           hipMemcpyHostToDevice, s.stream));
+  }
+  int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
+  for (int i = 0; i < n_src; i++) src_ids[i] = i;
+  for (int j = 0; j < n_out; j++) out_ids[j] = n_src + j;
+  const ecx::GF8& f = ecx::gf8();
+  for (int j0 = 0; j0 < n_out; j0 += 4) {
+    int nj = std::min(4, n_out - j0);
+    EcLaunchParams p;
+    std::vector<uint8_t> sub((size_t)nj * n_src);
+    for (int j = 0; j < nj; j++)
+      std::memcpy(&sub[(size_t)j * n_src], &rows[(size_t)(j0 + j) * n_src],
+                  n_src);
+    fill_params(&p, f, src_ids, n_src, out_ids + j0, nj, sub.data(),
+                src_null);
+    int rr = launch_matmul(ctx, s, s.d_stage, s.d_stage, p, 1, bytes, false,
+                           false);
+    if (rr != ECX_OK) return rr;
+  }
+  for (int j = 0; j < n_out; j++) {
+    if (!outs[j]) continue;
+    HIP_TRY(hipMemcpyAsync(outs[j], s.d_stage + (size_t)(n_src + j) * bytes,
+                           bytes, hipMemcpyDeviceToHost, s.stream));
+  }
+  HIP_TRY(hipStreamSynchronize(s.stream));
+  return ECX_OK;
+}
+
+int ecx_shec_matrix(int k, int m, int c, int single, uint8_t* out) {
+  if (!out) return ECX_ERR_INVAL;
+  std::vector<uint8_t> coding;
+  if (!ecx::shec_matrix(coding, k, m, c, single != 0)) return ECX_ERR_INVAL;
+  std::memcpy(out, coding.data(), coding.size());
+  return ECX_OK;
 }
 
 int ecx_sync(ecx_ctx* ctx, int slot) {

@@ -2,6 +2,8 @@
 #include "gf.h"
 
 #include <cstring>
+#include <algorithm>
+#include <limits>
 
 namespace ecx {
 
@@ -292,6 +294,248 @@ bool compose_decode_rows(const std::vector<uint8_t> &gen, int k, int m,
           s ^= f.mul(d[(size_t)j * k + i], gen[(size_t)e * k + j]);
         rows[p * k + i] = s;
       }
+    }
+  }
+  return true;
+}
+
+// ---- SHEC (restated from the reference's in-tree implementation) ----
+
+// shec_calc_recovery_efficiency1 (ErasureCodeShec.cc:660-697)
+static double shec_r_e1(int k, int m1, int m2, int c1, int c2) {
+  if (m1 < c1 || m2 < c2) return -1;
+  if ((m1 == 0 && c1 != 0) || (m2 == 0 && c2 != 0)) return -1;
+  std::vector<int> r_eff_k(k, 100000000);
+  double r_e1 = 0;
+  for (int rr = 0; rr < m1; rr++) {
+    int start = ((rr * k) / m1) % k;
+    int end = (((rr + c1) * k) / m1) % k;
+    int first = 1;
+    for (int cc = start; first || cc != end; cc = (cc + 1) % k) {
+      first = 0;
+      r_eff_k[cc] =
+          std::min(r_eff_k[cc], ((rr + c1) * k) / m1 - (rr * k) / m1);
+    }
+    r_e1 += ((rr + c1) * k) / m1 - (rr * k) / m1;
+  }
+  for (int rr = 0; rr < m2; rr++) {
+    int start = ((rr * k) / m2) % k;
+    int end = (((rr + c2) * k) / m2) % k;
+    int first = 1;
+    for (int cc = start; first || cc != end; cc = (cc + 1) % k) {
+      first = 0;
+      r_eff_k[cc] =
+          std::min(r_eff_k[cc], ((rr + c2) * k) / m2 - (rr * k) / m2);
+    }
+    r_e1 += ((rr + c2) * k) / m2 - (rr * k) / m2;
+  }
+  for (int i = 0; i < k; i++) r_e1 += r_eff_k[i];
+  return r_e1 / (k + m1 + m2);
+}
+
+// shec_reedsolomon_coding_matrix (ErasureCodeShec.cc:700-768), w=8
+bool shec_matrix(std::vector<uint8_t> &coding, int k, int m, int c,
+                 bool single) {
+  if (k < 1 || m < 1 || c < 1 || c > m) return false;
+  int m1, m2, c1, c2;
+  if (!single) {
+    int c1_best = -1, m1_best = -1;
+    double min_r_e1 = 100.0;
+    for (int tc1 = 0; tc1 <= c / 2; tc1++) {
+      for (int tm1 = 0; tm1 <= m; tm1++) {
+        int tc2 = c - tc1, tm2 = m - tm1;
+        if (tm1 < tc1 || tm2 < tc2) continue;
+        if ((tm1 == 0 && tc1 != 0) || (tm2 == 0 && tc2 != 0)) continue;
+        if ((tm1 != 0 && tc1 == 0) || (tm2 != 0 && tc2 == 0)) continue;
+        double r = shec_r_e1(k, tm1, tm2, tc1, tc2);
+        if (min_r_e1 - r > std::numeric_limits<double>::epsilon() &&
+            r < min_r_e1) {
+          min_r_e1 = r;
+          c1_best = tc1;
+          m1_best = tm1;
+        }
+      }
+    }
+    m1 = m1_best;
+    c1 = c1_best;
+    m2 = m - m1_best;
+    c2 = c - c1_best;
+    if (m1 < 0) return false;
+  } else {
+    m1 = 0; c1 = 0; m2 = m; c2 = c;
+  }
+  std::vector<uint8_t> full;
+  if (!gen_matrix_rs_van_jerasure(full, k, m)) return false;
+  coding.assign(full.begin() + (size_t)k * k, full.end());
+  for (int rr = 0; rr < m1; rr++) {
+    int end = ((rr * k) / m1) % k;
+    int start = (((rr + c1) * k) / m1) % k;
+    for (int cc = start; cc != end; cc = (cc + 1) % k)
+      coding[(size_t)rr * k + cc] = 0;
+  }
+  for (int rr = 0; rr < m2; rr++) {
+    int end = ((rr * k) / m2) % k;
+    int start = (((rr + c2) * k) / m2) % k;
+    for (int cc = start; cc != end; cc = (cc + 1) % k)
+      coding[(size_t)(rr + m1) * k + cc] = 0;
+  }
+  return true;
+}
+
+// GF(2^8) determinant via elimination (char 2: no sign bookkeeping);
+// semantics match shec determinant.c's zero/nonzero answer.
+static uint8_t gf_det(std::vector<uint8_t> mat, int n) {
+  const GF8 &f = gf8();
+  uint8_t det = 1;
+  for (int i = 0; i < n; i++) {
+    if (!mat[(size_t)i * n + i]) {
+      int j = i + 1;
+      while (j < n && !mat[(size_t)j * n + i]) j++;
+      if (j >= n) return 0;
+      for (int cc = 0; cc < n; cc++)
+        std::swap(mat[(size_t)i * n + cc], mat[(size_t)j * n + cc]);
+    }
+    uint8_t piv = mat[(size_t)i * n + i];
+    det = f.mul(det, piv);
+    uint8_t inv = f.inv(piv);
+    for (int r = i + 1; r < n; r++) {
+      uint8_t t = f.mul(inv, mat[(size_t)r * n + i]);
+      if (!t) continue;
+      for (int cc = i; cc < n; cc++)
+        mat[(size_t)r * n + cc] ^= f.mul(t, mat[(size_t)i * n + cc]);
+    }
+  }
+  return det;
+}
+
+// shec_make_decoding_matrix + shec_matrix_decode marshalling
+// (ErasureCodeShec.cc:770-1050). Returns false when unrecoverable.
+bool shec_decode_plan(const std::vector<uint8_t> &coding, int k, int m,
+                      uint64_t want_mask, uint64_t avail_mask,
+                      ShecPlan &plan) {
+  std::vector<int> want(k + m, 0), avails(k + m, 0);
+  for (int i = 0; i < k + m; i++) {
+    want[i] = (want_mask >> i) & 1;
+    avails[i] = (avail_mask >> i) & 1;
+  }
+  // wanting a lost parity pulls in the data chunks it covers (:782-790)
+  for (int i = 0; i < m; i++)
+    if (want[k + i] && !avails[k + i])
+      for (int j = 0; j < k; j++)
+        if (coding[(size_t)i * k + j] > 0) want[j] = 1;
+
+  int mindup = k + 1, minp = k + 1;
+  std::vector<int> best_row, best_col;
+  for (unsigned long long pp = 0; pp < (1ull << m); ++pp) {
+    int ek = 0;
+    std::vector<int> p(m);
+    for (int i = 0; i < m; i++)
+      if (pp & (1ull << i)) p[ek++] = i;
+    if (ek > minp) continue;
+    bool ok = true;
+    for (int i = 0; i < ek && ok; i++)
+      if (!avails[k + p[i]]) ok = false;
+    if (!ok) continue;
+
+    std::vector<int> tmprow(k + m, 0), tmpcol(k, 0);
+    for (int i = 0; i < k; i++)
+      if (want[i] && !avails[i]) tmpcol[i] = 1;
+    for (int i = 0; i < ek; i++) {
+      tmprow[k + p[i]] = 1;
+      for (int j = 0; j < k; j++) {
+        uint8_t el = coding[(size_t)p[i] * k + j];
+        if (el != 0) tmpcol[j] = 1;
+        if (el != 0 && avails[j] == 1) tmprow[j] = 1;
+      }
+    }
+    int dup_row = 0, dup_col = 0;
+    for (int i = 0; i < k + m; i++) dup_row += tmprow[i];
+    for (int i = 0; i < k; i++) dup_col += tmpcol[i];
+    if (dup_row != dup_col) continue;
+    int dup = dup_row;
+    if (dup == 0) {
+      mindup = 0;
+      best_row.clear();
+      best_col.clear();
+      break;
+    }
+    if (dup < mindup) {
+      std::vector<uint8_t> tmpmat((size_t)dup * dup);
+      int row = 0;
+      for (int i = 0; i < k + m; i++) {
+        if (!tmprow[i]) continue;
+        int col = 0;
+        for (int j = 0; j < k; j++) {
+          if (!tmpcol[j]) continue;
+          tmpmat[(size_t)row * dup + col] =
+              i < k ? (i == j ? 1 : 0) : coding[(size_t)(i - k) * k + j];
+          col++;
+        }
+        row++;
+      }
+      if (gf_det(tmpmat, dup) != 0) {
+        mindup = dup;
+        minp = ek;
+        best_row.clear();
+        best_col.clear();
+        for (int i = 0; i < k + m; i++)
+          if (tmprow[i]) best_row.push_back(i);
+        for (int i = 0; i < k; i++)
+          if (tmpcol[i]) best_col.push_back(i);
+      }
+    }
+  }
+  if (mindup == k + 1) return false;
+
+  // minimum set (:957-985)
+  plan.minimum.clear();
+  std::vector<int> minimum(k + m, 0);
+  for (int id : best_row) minimum[id] = 1;
+  for (int i = 0; i < k; i++)
+    if (want[i] && avails[i]) minimum[i] = 1;
+  for (int i = 0; i < m; i++) {
+    if (want[k + i] && avails[k + i] && !minimum[k + i]) {
+      for (int j = 0; j < k; j++) {
+        if (coding[(size_t)i * k + j] > 0 && !want[j]) {
+          minimum[k + i] = 1;
+          break;
+        }
+      }
+    }
+  }
+  for (int i = 0; i < k + m; i++)
+    if (minimum[i]) plan.minimum.push_back(i);
+
+  plan.src_ids.clear();
+  plan.out_ids.clear();
+  plan.rows.clear();
+  if (mindup > 0) {
+    std::vector<uint8_t> tmpmat((size_t)mindup * mindup),
+        inv((size_t)mindup * mindup);
+    for (int i = 0; i < mindup; i++)
+      for (int j = 0; j < mindup; j++)
+        tmpmat[(size_t)i * mindup + j] =
+            best_row[i] < k
+                ? (best_row[i] == best_col[j] ? 1 : 0)
+                : coding[(size_t)(best_row[i] - k) * k + best_col[j]];
+    if (!gf_invert(tmpmat.data(), inv.data(), mindup)) return false;
+    plan.src_ids = best_row;  // original chunk ids (data or parity)
+    for (int i = 0; i < mindup; i++) {
+      if (avails[best_col[i]]) continue;  // only erased columns computed
+      plan.out_ids.push_back(best_col[i]);
+      plan.rows.insert(plan.rows.end(), inv.begin() + (size_t)i * mindup,
+                       inv.begin() + (size_t)(i + 1) * mindup);
+    }
+  }
+  // phase 2: re-encode wanted lost parity from (recovered) data (:1040-1046)
+  plan.parity_out.clear();
+  plan.parity_rows.clear();
+  for (int i = 0; i < m; i++) {
+    if (want[k + i] && !avails[k + i]) {
+      plan.parity_out.push_back(k + i);
+      plan.parity_rows.insert(plan.parity_rows.end(),
+                              coding.begin() + (size_t)i * k,
+                              coding.begin() + (size_t)(i + 1) * k);
     }
   }
   return true;

@@ -74,4 +74,30 @@ bool compose_decode_rows(const std::vector<uint8_t> &gen, int k, int m,
                          std::vector<int> &erased,
                          std::vector<uint8_t> &rows);
 
+// SHEC (shingled EC) matrix + decode search, restated from the reference's
+// OWN in-tree implementation (src/erasure-code/shec/ErasureCodeShec.cc —
+// unlike the GF submodules this algorithm is fully present):
+//  - shec_matrix: shec_reedsolomon_coding_matrix (:700-768): RS-van
+//    jerasure matrix with shingle ranges zeroed; `single` selects the
+//    SINGLE vs MULTIPLE technique split (m1/c1 search via
+//    shec_calc_recovery_efficiency1, :660-697).
+//  - shec_decode_plan: shec_make_decoding_matrix (:770-...): minimal
+//    parity-subset search with determinant test, inversion, and the
+//    minimum set. Returns false when the pattern is unrecoverable.
+struct ShecPlan {
+  // phase 1: recover erased data chunks
+  std::vector<int> src_ids;   // chunk ids (data survivors + chosen parity)
+  std::vector<int> out_ids;   // erased data chunk ids to recover
+  std::vector<uint8_t> rows;  // out x src coefficients
+  // phase 2: re-encode erased wanted parity from (recovered) data
+  std::vector<int> parity_out;          // chunk ids >= k
+  std::vector<uint8_t> parity_rows;     // parity_out x k coefficients
+  std::vector<int> minimum;             // chunk ids to fetch
+};
+bool shec_matrix(std::vector<uint8_t> &coding /* m x k */, int k, int m,
+                 int c, bool single);
+bool shec_decode_plan(const std::vector<uint8_t> &coding, int k, int m,
+                      uint64_t want_mask, uint64_t avail_mask,
+                      ShecPlan &plan);
+
 }  // namespace ecx

@@ -184,6 +184,27 @@ ECX_API int ecx_decode_slices(ecx_ctx *ctx, void *const *d_chunks,
                               const size_t *bytes, int n_slices,
                               uint64_t present_mask, int slot);
 
+/* Replace the coding rows of the generator with a custom m x k matrix
+ * (clears decode-plan caches). Lets composed codecs — SHEC's shingled
+ * matrix (src/erasure-code/shec/ErasureCodeShec.cc:700-768), custom
+ * research codes — reuse every standard entry point. Matrix techniques
+ * only (not bitmatrix). */
+ECX_API int ecx_set_matrix(ecx_ctx *ctx, const uint8_t *coding_rows);
+
+/* Generic GF(2^8) matmul over host chunks: outs[j] = XOR_i rows[j*n_src+i]
+ * * srcs[i]. The primitive behind jerasure_matrix_dotprod
+ * (used by SHEC decode, ErasureCodeShec.cc:1030-1046); srcs may be NULL
+ * (zeros). Stages over PCIe and runs the standard kernel. */
+ECX_API int ecx_matmul_chunks_host(ecx_ctx *ctx,
+                                   const uint8_t *const *srcs, int n_src,
+                                   uint8_t *const *outs, int n_out,
+                                   const uint8_t *rows, size_t bytes);
+
+/* SHEC shingled coding matrix (m x k), restated from the reference's
+ * in-tree shec_reedsolomon_coding_matrix (ErasureCodeShec.cc:700-768);
+ * single != 0 selects the SINGLE technique. */
+ECX_API int ecx_shec_matrix(int k, int m, int c, int single, uint8_t *out);
+
 /* Synchronise a stream slot. */
 ECX_API int ecx_sync(ecx_ctx *ctx, int slot);
 

@@ -126,6 +126,21 @@ def encode(technique, k, m, data, chunk_bytes=None):
     return parity
 
 
+def encode_with_rows(rows, data, chunk_bytes=None):
+    """Encode with an explicit m x k coding matrix (e.g. a SHEC shingled
+    matrix) — the checker for custom-matrix codecs."""
+    m, k = rows.shape
+    lens = {d.nbytes for d in data if d is not None}
+    assert len(lens) == 1 or (not lens and chunk_bytes)
+    length = lens.pop() if lens else chunk_bytes
+    rows = np.ascontiguousarray(rows)
+    parity = [np.zeros(length, dtype=np.uint8) for _ in range(m)]
+    _ref.ecref_encode(k, m, rows.ctypes.data_as(ctypes.c_void_p),
+                      _ptr_array(data), _ptr_array(parity),
+                      ctypes.c_size_t(length))
+    return parity
+
+
 def decode(technique, k, m, chunks, present):
     """chunks: list of k+m uint8 arrays (erased entries are overwritten in
     place with the reconstruction). present: list/array of 0/1 flags."""
