@@ -23,6 +23,10 @@ int technique_id(const std::string &t) {
   if (t == "cauchy") return ECX_T_CAUCHY_ISA;
   if (t == "jerasure_reed_sol_van") return ECX_T_RS_VAN_JERASURE;
   if (t == "cauchy_orig") return ECX_T_CAUCHY_ORIG_JERASURE;
+  // jerasure reed_sol_r6_op (RAID6, m==2): its reed_sol_r6_coding_matrix
+  // is row0 = all ones, row1 = [2^j] — byte-identical to the isa RS-van
+  // construction at m=2 (gf_gen_rs_matrix rows k, k+1)
+  if (t == "reed_sol_r6_op") return ECX_T_RS_VAN_ISA;
   return -1;
 }
 
@@ -75,6 +79,11 @@ class ErasureCodeMi355x final : public ErasureCode {
     }
     if (technique_id(technique_) < 0) {
       if (ss) *ss << "mi355x: unknown technique " << technique_ << "\n";
+      err = -EINVAL;
+    }
+    if (technique_ == "reed_sol_r6_op" && m_ != 2) {
+      // ErasureCodeJerasureReedSolomonRAID6::parse (:473-488)
+      if (ss) *ss << "reed_sol_r6_op: m=" << m_ << " must be 2 for RAID6\n";
       err = -EINVAL;
     }
     profile["technique"] = technique_;

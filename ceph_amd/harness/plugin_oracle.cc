@@ -24,6 +24,7 @@ int technique_id(const std::string &t) {
   if (t == "cauchy") return ECREF_T_CAUCHY_ISA;
   if (t == "jerasure_reed_sol_van") return ECREF_T_RS_VAN_JERASURE;
   if (t == "cauchy_orig") return 3;  // bitmatrix/packet layout
+  if (t == "reed_sol_r6_op") return ECREF_T_RS_VAN_ISA;  // RAID6 == isa m=2
   return -1;
 }
 
@@ -56,6 +57,10 @@ class ErasureCodeOracle final : public ErasureCode {
     int t = technique_id(technique_);
     if (t < 0) {
       if (ss) *ss << "oracle: unknown technique " << technique_ << "\n";
+      err = -EINVAL;
+    }
+    if (technique_ == "reed_sol_r6_op" && m_ != 2) {
+      if (ss) *ss << "reed_sol_r6_op: m must be 2\n";
       err = -EINVAL;
     }
     if (is_bitmatrix())

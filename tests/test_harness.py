@@ -83,3 +83,16 @@ def test_mi355x_plugin_loads_without_gpu():
     compute init refuses without a GPU (covered on the GPU box)."""
     lib = ctypes.CDLL(os.path.join(HARNESS, "libec_mi355x.so"))
     assert lib is not None
+
+
+def test_reed_sol_r6_technique():
+    """jerasure reed_sol_r6_op (RAID6): matrix == isa RS-van at m=2
+    (all-ones row + powers-of-2 row); m != 2 rejected
+    (ErasureCodeJerasure.cc:473-488)."""
+    r = run_bench("-p", "oracle", "-P", "technique=reed_sol_r6_op",
+                  "-P", "k=5", "-P", "m=2", "-s", "65536", "-i", "2",
+                  "-w", "decode", "-e", "2", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr
+    r = run_bench("-p", "oracle", "-P", "technique=reed_sol_r6_op",
+                  "-P", "k=5", "-P", "m=3", "-s", "65536", "-i", "1")
+    assert r.returncode != 0
