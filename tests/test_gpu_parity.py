@@ -305,3 +305,30 @@ def test_cauchy_orig_batch_roundtrip():
         assert np.array_equal(out, ref)
     finally:
         ctx.close()
+
+
+@pytest.mark.parametrize("k,m", [(16, 6), (32, 4), (20, 8)])
+def test_large_km_parity(k, m):
+    """Upper end of the isa limits (MAX_K=MAX_M=32, ErasureCodeIsa.h:48):
+    wide stripes exercise the n_out>4 launch splitting and 64-bit masks."""
+    C = 16 * 1024
+    tech = "cauchy"  # Vandermonde is MDS-limited to m<=4 (ErasureCodeIsa.cc:598-631)
+    rng = np.random.default_rng(k * 100 + m)
+    ctx = make_ctx(k, m, tech)
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        got = ctx.encode_chunks(data)
+        want = oracle.encode(tech, k, m, data)
+        for j in range(m):
+            assert np.array_equal(got[j], want[j]), j
+        # decode with m/2 random erasures
+        full = data + got
+        er = sorted(rng.choice(k + m, size=m // 2, replace=False).tolist())
+        present = [i not in er for i in range(k + m)]
+        chunks = [c.copy() if present[i] else np.zeros(C, np.uint8)
+                  for i, c in enumerate(full)]
+        ctx.decode_chunks(chunks, present)
+        for i in range(k + m):
+            assert np.array_equal(chunks[i], full[i]), (er, i)
+    finally:
+        ctx.close()
