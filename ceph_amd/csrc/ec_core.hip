@@ -936,11 +936,12 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   }();
 
   // LDS window: largest power-of-two divisor of pkt within the LDS budget
-  // (ECX_BITQ KB, default 48 — A/B-able; bigger windows fill waves better
-  // but cost residency)
+  // (ECX_BITQ KB). Default 16 from the MI355X sweep: small windows keep
+  // 8+ blocks/CU resident and beat large windows by ~19%
+  // (profiles/rocprof_r01_summary.md).
   static const size_t lds_budget = [] {
     const char* v = getenv("ECX_BITQ");
-    long kb = v ? atol(v) : 48;
+    long kb = v ? atol(v) : 16;
     if (kb < 8) kb = 8;
     if (kb > 120) kb = 120;
     return (size_t)kb * 1024;
