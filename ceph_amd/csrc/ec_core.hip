@@ -30,6 +30,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <cstdint>
 #include <algorithm>
 #include <cstring>
@@ -443,6 +444,10 @@ struct ecx_ctx {
   std::list<uint64_t> lru_order;
   std::map<uint64_t, BitPlan> bit_lru;  // bitmatrix decode plans
   static constexpr size_t LRU_DEPTH = 4096;
+  // host-pointer calls round-robin the stream slots so concurrent plugin
+  // threads (the OSD's PG workers) overlap instead of serialising on one
+  // stream's mutex
+  std::atomic<unsigned> rr{0};
 
   bool is_bitmatrix() const { return technique == ECX_T_CAUCHY_ORIG_JERASURE; }
 };
@@ -1173,7 +1178,7 @@ int ecx_matmul_chunks_host(ecx_ctx* ctx, const uint8_t* const* srcs,
   if (!ctx || !srcs || !outs || !rows || n_src < 1 || n_src > ECX_MAX_K ||
       n_out < 1 || n_out > ECX_MAX_K || bytes % 16)
     return ECX_ERR_INVAL;
-  Slot& s = ctx->slots[0];
+  Slot& s = ctx->slots[ctx->rr++ % ctx->slots.size()];
   std::lock_guard<std::recursive_mutex> g(s.mu);
   int r = ensure_stage(ctx, s, (size_t)(n_src + n_out) * bytes);
   if (r != ECX_OK) return r;
@@ -1261,7 +1266,8 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
                            uint8_t* const* parity, size_t chunk_bytes) {
   if (!ctx || !data || !parity || chunk_bytes % 16) return ECX_ERR_INVAL;
   int k = ctx->k, m = ctx->m;
-  Slot& s = ctx->slots[0];
+  const int si = (int)(ctx->rr++ % ctx->slots.size());
+  Slot& s = ctx->slots[si];
   std::lock_guard<std::recursive_mutex> g(s.mu);
   int r = ensure_stage(ctx, s, (size_t)(k + m) * chunk_bytes);
   if (r != ECX_OK) return r;
@@ -1287,7 +1293,7 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
     for (int j0 = 0; j0 < m; j0 += ECX_MAX_OUT) {
       int nj = std::min(ECX_MAX_OUT, m - j0);
       int rr = run_bitmatrix(
-          ctx, 0, s.d_stage, s.d_stage, src_ids, k, out_ids + j0, nj,
+          ctx, si, s.d_stage, s.d_stage, src_ids, k, out_ids + j0, nj,
           ctx->bitmat.data() + (size_t)(j0 * ctx->w) * k * ctx->w, 1,
           chunk_bytes, false);
       if (rr != ECX_OK) return rr;
@@ -1326,7 +1332,8 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
     int r = get_bit_plan(ctx, present_mask, plan);
     if (r != ECX_OK) return r;
     if (plan.erased.empty()) return ECX_OK;
-    Slot& s = ctx->slots[0];
+    const int si = (int)(ctx->rr++ % ctx->slots.size());
+    Slot& s = ctx->slots[si];
     std::lock_guard<std::recursive_mutex> g(s.mu);
     r = ensure_stage(ctx, s, (size_t)n * chunk_bytes);
     if (r != ECX_OK) return r;
@@ -1345,7 +1352,7 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
     for (size_t j0 = 0; j0 < plan.erased.size(); j0 += ECX_MAX_OUT) {
       int nj = (int)std::min<size_t>(ECX_MAX_OUT, plan.erased.size() - j0);
       int rr = run_bitmatrix(
-          ctx, 0, s.d_stage, s.d_stage, plan.survivors.data(), k,
+          ctx, si, s.d_stage, s.d_stage, plan.survivors.data(), k,
           plan.erased.data() + j0, nj,
           plan.rows.data() + j0 * (size_t)ctx->w * k * ctx->w, 1,
           chunk_bytes, false);
@@ -1365,7 +1372,7 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
   if (r != ECX_OK) return r;
   if (plan.erased.empty()) return ECX_OK;
 
-  Slot& s = ctx->slots[0];
+  Slot& s = ctx->slots[ctx->rr++ % ctx->slots.size()];
   std::lock_guard<std::recursive_mutex> g(s.mu);
   r = ensure_stage(ctx, s, (size_t)n * chunk_bytes);
   if (r != ECX_OK) return r;
@@ -1408,7 +1415,8 @@ int ecx_encode_delta_host(ecx_ctx* ctx, const uint8_t* old_data,
                           size_t bytes) {
   if (!ctx || !old_data || !new_data || !delta || bytes % 16)
     return ECX_ERR_INVAL;
-  Slot& s = ctx->slots[0];
+  const int si = (int)(ctx->rr++ % ctx->slots.size());
+  Slot& s = ctx->slots[si];
   std::lock_guard<std::recursive_mutex> g(s.mu);
   int r = ensure_stage(ctx, s, 3 * bytes);
   if (r != ECX_OK) return r;
@@ -1418,7 +1426,7 @@ int ecx_encode_delta_host(ecx_ctx* ctx, const uint8_t* old_data,
   HIP_TRY(hipMemcpyAsync(s.d_stage + bytes, new_data, bytes,
                          hipMemcpyHostToDevice, s.stream));
   r = ecx_encode_delta_dev(ctx, s.d_stage, s.d_stage + bytes,
-                           s.d_stage + 2 * bytes, bytes, 0);
+                           s.d_stage + 2 * bytes, bytes, si);
   if (r != ECX_OK) return r;
   HIP_TRY(hipMemcpyAsync(delta, s.d_stage + 2 * bytes, bytes,
                          hipMemcpyDeviceToHost, s.stream));
@@ -1429,7 +1437,8 @@ int ecx_encode_delta_host(ecx_ctx* ctx, const uint8_t* old_data,
 int ecx_apply_delta_host(ecx_ctx* ctx, const uint8_t* delta, int data_shard,
                          int coding_shard, uint8_t* parity, size_t bytes) {
   if (!ctx || !delta || !parity || bytes % 16) return ECX_ERR_INVAL;
-  Slot& s = ctx->slots[0];
+  const int si = (int)(ctx->rr++ % ctx->slots.size());
+  Slot& s = ctx->slots[si];
   std::lock_guard<std::recursive_mutex> g(s.mu);
   int r = ensure_stage(ctx, s, 2 * bytes);
   if (r != ECX_OK) return r;
@@ -1439,7 +1448,7 @@ int ecx_apply_delta_host(ecx_ctx* ctx, const uint8_t* delta, int data_shard,
   HIP_TRY(hipMemcpyAsync(s.d_stage + bytes, parity, bytes,
                          hipMemcpyHostToDevice, s.stream));
   r = ecx_apply_delta_dev(ctx, s.d_stage, data_shard, coding_shard,
-                          s.d_stage + bytes, bytes, 0);
+                          s.d_stage + bytes, bytes, si);
   if (r != ECX_OK) return r;
   HIP_TRY(hipMemcpyAsync(parity, s.d_stage + bytes, bytes,
                          hipMemcpyDeviceToHost, s.stream));
