@@ -306,11 +306,13 @@ def main():
     for e in erased:
         present_mask &= ~(1 << e)
 
-    def step(timed_accum=None):
+    def step(timed_accum=None, dec_accum=None):
         ctx.encode_batch(dptr, S, C)
         if timed_accum is not None:
             timed_accum.append(ctx.last_kernel_ms())
         ctx.decode_batch(dptr, S, C, present_mask)
+        if dec_accum is not None:
+            dec_accum.append(ctx.last_kernel_ms())
         ctx.sync()
 
     # parity self-check before measuring (rank 0)
@@ -325,10 +327,10 @@ def main():
         import torch
         torch.cuda.synchronize()
 
-    enc_ms = []
+    enc_ms, dec_ms = [], []
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step(enc_ms)
+        step(enc_ms, dec_ms)
     elapsed = time.perf_counter() - t0
     if dist:
         import torch
@@ -380,9 +382,12 @@ def main():
                 "frac": round(achieved / peak, 4) if achieved else None,
                 "traffic": None,  # PMC traffic comes from rocprofv3 runs
                                   # committed under profiles/
-                "kernel": "ec_gf_matmul_kernel<3,false> (encode)",
+                "kernel": "ec_gf_matmul_kernel (encode leg)",
                 "kernel_ms": round(enc_kernel_ms, 4) if enc_kernel_ms else None,
                 "alg_bytes_per_launch": alg_bytes,
+                "decode_kernel_ms": (round(float(np.mean(dec_ms)), 4)
+                                     if dec_ms else None),
+                "decode_alg_bytes": (k + len(erased)) * C * S,
             },
             "cpu_baseline": (cpu_baseline(args)
                              if (world == 1 and not args.no_cpu_baseline
