@@ -23,7 +23,7 @@ ceph_amd/libec_mi355x_core.so: ceph_amd/csrc/ec_core.hip ceph_amd/csrc/gf.cpp ce
 HARNESS_HDRS = $(HARNESS)/ec_types.h $(HARNESS)/erasure_code.h $(HARNESS)/erasure_code_plugin.h
 
 harness: $(HARNESS)/ec_benchmark $(HARNESS)/registry_selftest \
-         $(HARNESS)/libec_mi355x.so $(HARNESS)/libec_oracle.so $(HARNESS)/libec_lrc.so $(HARNESS)/libec_shec.so \
+         $(HARNESS)/libec_mi355x.so $(HARNESS)/libec_oracle.so $(HARNESS)/libec_lrc.so $(HARNESS)/libec_shec.so $(HARNESS)/libec_clay.so \
          $(HARNESS)/libec_fix_missing_version.so $(HARNESS)/libec_fix_bad_version.so \
          $(HARNESS)/libec_fix_missing_init.so $(HARNESS)/libec_fix_fail_init.so \
          $(HARNESS)/libec_fix_no_register.so
@@ -51,6 +51,10 @@ $(HARNESS)/libec_oracle.so: $(HARNESS)/plugin_oracle.cc oracle/ec_ref.c oracle/e
 $(HARNESS)/libec_shec.so: $(HARNESS)/plugin_shec.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc ceph_amd/libec_mi355x_core.so $(HARNESS_HDRS)
 	$(CXX) $(CXXFLAGS) -shared $(HARNESS)/plugin_shec.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc \
 	  -L ceph_amd -lec_mi355x_core '-Wl,-rpath,$$ORIGIN/..' -L $(ROCM)/lib -lamdhip64 -ldl -o $@
+
+# Clay coupled-layer plugin (composes registry sub-codecs)
+$(HARNESS)/libec_clay.so: $(HARNESS)/plugin_clay.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc $(HARNESS_HDRS)
+	$(CXX) $(CXXFLAGS) -shared $(HARNESS)/plugin_clay.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc -ldl -o $@
 
 # LRC layered-composition plugin (host-side dispatch over registry
 # sub-plugins; default sub-plugin mi355x, 'oracle' for CPU tests)
