@@ -133,7 +133,7 @@ def bench_mixed(args):
     shapes = [(4, 2, 64 << 10), (6, 3, 256 << 10), (8, 3, 1 << 20),
               (12, 4, 4 << 20)]
     rng = np.random.default_rng(args.seed)
-    ctxs, bufs = [], []
+    ctxs = []
     for (k, m, C) in shapes:
         S = max(8, int(2 * GIB // (k * C)))
         ctx = ceph_amd.EcContext(k, m, "reed_sol_van", device=local_rank,
