@@ -11,9 +11,11 @@
 //
 // v1 scope: encode/decode via the layered path (bit-correct, any <= m
 // erasures). The bandwidth-optimal single-node repair path
-// (repair_one_lost_chunk :522-700, minimum_to_repair sub-chunk lists) is a
-// later round; minimum_to_decode returns full sub-chunk ranges, which is
-// correct but not repair-bandwidth-optimal.
+// (repair_one_lost_chunk :522-700) is NOT wired — matching the reference's
+// CURRENT behaviour: its is_repair() unconditionally returns 0 ("XXX: for
+// now returning 0 and knocking the optimization out",
+// ErasureCodeClay.cc:357-370), so upstream clay also always takes the full
+// decode path today. minimum_to_decode returns full sub-chunk ranges.
 //
 // Deviations from the reference, deliberate: modern shard_id API (the
 // reference clay still uses the deprecated std::set<int> forms); U_buf is
