@@ -84,3 +84,13 @@ def test_clay_gpu_matches_cpu_oracle_subcodec():
                    "-s", str(6 * 64 * 1024), "-i", "1", "-w", "decode",
                    "-e", "3", "-E", "exhaustive")
     assert r1.returncode == 0, r1.stderr + r1.stdout
+
+
+def test_chunk_size_alignment_rule():
+    """get_chunk_size = stripe rounded up to sub_chunk_no*k*scalar_align,
+    / k (ErasureCodeClay.cc:96-103): chunks must be sub-chunk divisible.
+    Verified through the CLI round trip at an awkward stripe width."""
+    r = run_bench(*CPU, "-P", "k=4", "-P", "m=2", "-P", "d=5",
+                  "-s", "10000", "-i", "1", "-w", "decode", "-e", "1",
+                  "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout

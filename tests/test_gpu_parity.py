@@ -332,3 +332,31 @@ def test_large_km_parity(k, m):
             assert np.array_equal(chunks[i], full[i]), (er, i)
     finally:
         ctx.close()
+
+
+def test_decode_insufficient_chunks_fails_loudly():
+    """Fewer than k survivors must fail with EIO, not fabricate data
+    (ErasureCodeInterface.h:29-35 error conventions)."""
+    k, m = 4, 2
+    C = 4096
+    ctx = make_ctx(k, m, "reed_sol_van")
+    try:
+        chunks = [np.zeros(C, np.uint8) for _ in range(k + m)]
+        present = [True, True, True, False, False, False]  # 3 < k
+        with pytest.raises(ceph_amd.EcError, match="EIO|too many"):
+            ctx.decode_chunks(chunks, present)
+    finally:
+        ctx.close()
+
+
+def test_minimum_to_decode_semantics():
+    ctx = make_ctx(6, 3, "reed_sol_van")
+    try:
+        want = 0b000001
+        avail = 0b111111111
+        assert ctx.minimum_to_decode(want, avail) == want
+        # chunk 0 lost: first k available in id order (ErasureCode.cc:154-170)
+        avail = 0b111111110
+        assert ctx.minimum_to_decode(want, avail) == 0b001111110
+    finally:
+        ctx.close()
