@@ -1237,6 +1237,7 @@ int ecx_last_kernel_ms(ecx_ctx* ctx, int slot, double* ms) {
   if (!ctx || !ms || slot < 0 || slot >= (int)ctx->slots.size())
     return ECX_ERR_INVAL;
   Slot& s = ctx->slots[slot];
+  std::lock_guard<std::recursive_mutex> g(s.mu);
   if (!s.timed) return ECX_ERR_INVAL;
   HIP_TRY(hipEventSynchronize(s.ev_stop));
   float f = 0.f;
