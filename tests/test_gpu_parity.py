@@ -360,3 +360,37 @@ def test_minimum_to_decode_semantics():
         assert ctx.minimum_to_decode(want, avail) == 0b001111110
     finally:
         ctx.close()
+
+
+def test_randomized_config_sweep():
+    """Seeded fuzz over (technique, k, m, C, erasure pattern): every config
+    bit-exact vs the oracle. One test, many configs — the long tail the
+    parametrized cases miss."""
+    rng = np.random.default_rng(0xF00D)
+    for trial in range(18):
+        tech = ["reed_sol_van", "cauchy",
+                "jerasure_reed_sol_van"][trial % 3]
+        k = int(rng.integers(2, 13))
+        m = int(rng.integers(1, 5))
+        if tech != "cauchy":
+            m = min(m, 4)  # Vandermonde MDS bound (ErasureCodeIsa.cc:598)
+        C = int(rng.integers(1, 64)) * 16
+        ctx = make_ctx(k, m, tech)
+        try:
+            data = [rng.integers(0, 256, C, dtype=np.uint8)
+                    for _ in range(k)]
+            got = ctx.encode_chunks(data)
+            want = oracle.encode(tech, k, m, data)
+            for j in range(m):
+                assert np.array_equal(got[j], want[j]), (trial, tech, k, m, C)
+            e = int(rng.integers(1, m + 1))
+            er = rng.choice(k + m, size=e, replace=False)
+            full = data + got
+            present = [i not in er for i in range(k + m)]
+            chunks = [c.copy() if present[i] else np.zeros(C, np.uint8)
+                      for i, c in enumerate(full)]
+            ctx.decode_chunks(chunks, present)
+            for i in range(k + m):
+                assert np.array_equal(chunks[i], full[i]), (trial, i)
+        finally:
+            ctx.close()
