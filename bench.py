@@ -79,9 +79,21 @@ def parity_selfcheck(ctx, dptr, args, seed, sample_stripes=2):
                     f"stripe {s}: GPU parity {j} mismatches oracle")
 
 
-def cpu_baseline(args, budget_s=12.0):
+def _set_omp_threads(n):
+    import ctypes
+    try:
+        ctypes.CDLL("libgomp.so.1").omp_set_num_threads(int(n))
+        return True
+    except OSError:
+        return False
+
+
+def cpu_baseline(args, budget_s=10.0):
     """Time the oracle's ISA-L-class AVX2/OpenMP path (kind='port') on a
-    bounded sample of the same workload on this host's cores."""
+    bounded sample of the same workload on this host's cores. Thread count
+    is CALIBRATED: GPU-box hosts expose 256 hardware threads but collapse
+    under full oversubscription (224 GiB/s at 64 threads vs 7 GiB/s at
+    256, measured), so we pick the best of a few counts and report it."""
     import oracle
     k, m, C = args.k, args.m, args.chunk_bytes
     S = max(1, min(args.stripes, int(2 * GIB / ((k + m) * C))))  # <=2 GiB data
@@ -91,8 +103,21 @@ def cpu_baseline(args, budget_s=12.0):
     present = np.ones(k + m, np.uint8)
     present[sorted(np.random.default_rng(0xEC).choice(
         k + m, size=args.erasures, replace=False))] = 0
-    # warm
-    oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)
+
+    hw = len(os.sched_getaffinity(0))
+    candidates = sorted({min(hw, n) for n in (8, 16, 32, 64, 96)})
+    best_nt, best = candidates[0], 0.0
+    oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)  # warm
+    for nt in candidates:
+        if not _set_omp_threads(nt):
+            break
+        t0 = time.perf_counter()
+        oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)
+        r = 1.0 / (time.perf_counter() - t0)
+        if r > best:
+            best, best_nt = r, nt
+    _set_omp_threads(best_nt)
+
     t0 = time.perf_counter()
     iters = 0
     while time.perf_counter() - t0 < budget_s:
@@ -104,10 +129,11 @@ def cpu_baseline(args, budget_s=12.0):
     return {
         "value": round(gibs, 3),
         "unit": "GiB/s",
-        "cores": oracle.cpu_threads(),
+        "cores": best_nt,
         "kind": "port",
         "sample": (f"{iters}x encode+decode of a {S}-stripe batch "
-                   f"(k={k},m={m},C={C}) in {dt:.1f}s on host cores"),
+                   f"(k={k},m={m},C={C}) in {dt:.1f}s; thread count "
+                   f"calibrated over {candidates} on a {hw}-thread host"),
     }
 
 

@@ -80,7 +80,9 @@ __device__ __forceinline__ uint32_t ecx_lut(uint32_t t0, uint32_t t1,
 // parallelism, half the loop overhead); both vectors of a lane are
 // blockDim.x apart so every wave access stays a fully coalesced 1 KiB
 // transaction.
-template <int NOUT, bool ACCUM, int VPT, bool NT>
+// KN > 0 statically unrolls the source loop (flagship k=8): all k loads
+// are grouped ahead of the compute within a position, deepening MLP.
+template <int NOUT, bool ACCUM, int VPT, bool NT, int KN = 0>
 __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const EcLaunchParams* __restrict__ pb, long chunk_bytes,
@@ -89,7 +91,7 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
   __shared__ int s_src[ECX_MAX_K];
   __shared__ int s_out[ECX_MAX_OUT];
   __shared__ uint8_t s_cls[ECX_MAX_OUT * ECX_MAX_K];
-  const int n_src = pb->n_src;
+  const int n_src = KN ? KN : pb->n_src;
   for (int t = threadIdx.x; t < NOUT * n_src * 6; t += blockDim.x)
     s_tabs[t] = pb->tabs[t];
   for (int t = threadIdx.x; t < n_src; t += blockDim.x)
@@ -129,6 +131,7 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
         }
       }
 
+#pragma unroll
     for (int i = 0; i < n_src; i++) {
       const uint8_t* sp = sbase + (long)s_src[i] * chunk_bytes;
       uint32_t dq[VPT][4];
@@ -736,13 +739,25 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
 
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
   const int cps = ctx->k + ctx->m;
+  static const int env_k8 = [] {
+    const char* v = getenv("ECX_K8");
+    return v ? atoi(v) : 1;
+  }();
+  const bool k8 = env_k8 && !accum && vpt == 1 && params.n_src == 8;
 #define ECX_LAUNCH(NO, AC, VP, NTF)                                          \
   hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF>), grid,           \
                      dim3(256), 0, s.stream, d_buf, d_obuf, s.d_params,      \
                      (long)chunk_bytes, cps, vecs)
+#define ECX_LAUNCH_K8(NO, NTF)                                               \
+  hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, false, 1, NTF, 8>), grid,      \
+                     dim3(256), 0, s.stream, d_buf, d_obuf, s.d_params,      \
+                     (long)chunk_bytes, cps, vecs)
 #define ECX_VARIANT(NO, AC)                          \
   do {                                               \
-    if (vpt == 2) {                                  \
+    if (k8) {                                        \
+      if (env_nt) ECX_LAUNCH_K8(NO, true);           \
+      else ECX_LAUNCH_K8(NO, false);                 \
+    } else if (vpt == 2) {                           \
       if (env_nt) ECX_LAUNCH(NO, AC, 2, true);       \
       else ECX_LAUNCH(NO, AC, 2, false);             \
     } else {                                         \
@@ -765,6 +780,7 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
   }
 #undef ECX_DISPATCH
 #undef ECX_VARIANT
+#undef ECX_LAUNCH_K8
 #undef ECX_LAUNCH
   HIP_TRY(hipGetLastError());
   if (time_it) {
