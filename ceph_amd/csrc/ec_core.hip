@@ -131,7 +131,7 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
         }
       }
 
-#pragma unroll
+#pragma unroll (KN ? KN : 1)
     for (int i = 0; i < n_src; i++) {
       const uint8_t* sp = sbase + (long)s_src[i] * chunk_bytes;
       uint32_t dq[VPT][4];
