@@ -38,6 +38,7 @@ struct Bench {
   int run();
   int encode();
   int decode();
+  int flags();
   int decode_erasures(const shard_id_map<buffer> &all,
                       const shard_id_map<buffer> &chunks, int shard,
                       unsigned want_erasures,
@@ -59,6 +60,7 @@ int Bench::setup(int argc, char **argv) {
       {"parameter", required_argument, nullptr, 'P'},
       {"directory", required_argument, nullptr, 'd'},
       {"seed", required_argument, nullptr, 1001},
+      {"flags", no_argument, nullptr, 1002},
       {nullptr, 0, nullptr, 0}};
   int c;
   while ((c = getopt_long(argc, argv, "hvs:i:p:w:e:E:P:d:", longopts,
@@ -80,6 +82,7 @@ int Bench::setup(int argc, char **argv) {
       case 'd': directory = optarg; break;
       case 1000: erased.push_back(atoi(optarg)); break;
       case 1001: seed = strtoull(optarg, nullptr, 0); break;
+      case 1002: workload = "flags"; break;
       case 'P': {
         std::string s(optarg);
         auto eq = s.find('=');
@@ -115,9 +118,41 @@ buffer Bench::make_input() const {
   return in;
 }
 
+// --flags: factory the plugin and print its claimed optimization flags
+// (names per ErasureCodeInterface.h:694-709) so conformance tests can
+// assert claims == verified behaviour.
+int Bench::flags() {
+  auto &instance = ErasureCodePluginRegistry::instance();
+  ErasureCodeInterfaceRef erasure_code;
+  std::stringstream messages;
+  int code =
+      instance.factory(plugin, directory, profile, &erasure_code, &messages);
+  if (code) {
+    std::cerr << messages.str() << std::endl;
+    return code;
+  }
+  auto f = erasure_code->get_supported_optimizations();
+  static const std::pair<uint64_t, const char *> names[] = {
+      {1 << 0, "partialread"},    {1 << 1, "partialwrite"},
+      {1 << 2, "zeroinout"},      {1 << 3, "zeropadding"},
+      {1 << 4, "paritydelta"},    {1 << 5, "requiresubchunks"},
+      {1 << 6, "optimizedsupport"}, {1 << 7, "crcencodedecode"},
+      {1 << 8, "directreads"}};
+  bool first = true;
+  for (auto &[bit, name] : names) {
+    if (f & bit) {
+      std::cout << (first ? "" : ",") << name;
+      first = false;
+    }
+  }
+  std::cout << std::endl;
+  return 0;
+}
+
 int Bench::run() {
   ErasureCodePluginRegistry::instance().disable_dlclose = true;
   if (workload == "encode") return encode();
+  if (workload == "flags") return flags();
   return decode();
 }
 

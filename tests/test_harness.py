@@ -96,3 +96,21 @@ def test_reed_sol_r6_technique():
     r = run_bench("-p", "oracle", "-P", "technique=reed_sol_r6_op",
                   "-P", "k=5", "-P", "m=3", "-s", "65536", "-i", "1")
     assert r.returncode != 0
+
+
+@pytest.mark.parametrize("plugin,profile,expected", [
+    ("oracle", ["-P", "k=4", "-P", "m=2"],
+     "partialread,partialwrite,zeroinout,paritydelta,optimizedsupport"),
+    ("lrc", ["-P", "k=4", "-P", "m=2", "-P", "l=3",
+             "-P", "lrc-default-plugin=oracle"],
+     "partialread,partialwrite,zeroinout"),
+])
+def test_claimed_flags_match_verified_behaviour(plugin, profile, expected):
+    """Optimization-flag conformance (TestErasureCodePlugins.cc principle:
+    claim only what the tests verify). Behaviours behind each claimed flag
+    are covered elsewhere in the suite: zeroinout (test_zero_in_zero_out),
+    paritydelta (test_parity_delta_equivalence + registry_selftest),
+    partialread/systematic (registry_selftest systematic-prefix)."""
+    r = run_bench("-p", plugin, *profile, "--flags")
+    assert r.returncode == 0, r.stderr
+    assert r.stdout.strip() == expected
