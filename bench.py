@@ -105,15 +105,17 @@ def cpu_baseline(args, budget_s=10.0):
         k + m, size=args.erasures, replace=False))] = 0
 
     hw = len(os.sched_getaffinity(0))
-    candidates = sorted({min(hw, n) for n in (8, 16, 32, 64, 96)})
+    candidates = sorted({min(hw, n) for n in (8, 16, 32, 64)})
     best_nt, best = candidates[0], 0.0
     oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)  # warm
     for nt in candidates:
         if not _set_omp_threads(nt):
             break
-        t0 = time.perf_counter()
-        oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)
-        r = 1.0 / (time.perf_counter() - t0)
+        r = 0.0
+        for _ in range(2):  # best-of-2: single shots are noisy
+            t0 = time.perf_counter()
+            oracle.cpu_encode_batch(args.technique, k, m, batch, S, C)
+            r = max(r, 1.0 / (time.perf_counter() - t0))
         if r > best:
             best, best_nt = r, nt
     _set_omp_threads(best_nt)

@@ -739,9 +739,13 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
 
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
   const int cps = ctx->k + ctx->m;
+  // K8 full source-unroll measured 2.1x SLOWER (16.7 vs 8.0 ms encode:
+  // the unrolled body bloats registers/issue and the dynamic loop already
+  // gets enough MLP from 8 waves/SIMD) — keep available for experiments,
+  // default OFF.
   static const int env_k8 = [] {
     const char* v = getenv("ECX_K8");
-    return v ? atoi(v) : 1;
+    return v ? atoi(v) : 0;
   }();
   const bool k8 = env_k8 && !accum && vpt == 1 && params.n_src == 8;
 #define ECX_LAUNCH(NO, AC, VP, NTF)                                          \
