@@ -65,14 +65,19 @@ struct EcLaunchParams {
 
 typedef uint32_t v4u __attribute__((ext_vector_type(4)));
 
-// 16-entry byte lookup over 4 packed bytes:
-//   T7 entries t0 (0..3) / t1 (4..7) selected by i7 in 0..7,
-//   2-entry T8 (byte1 of t8) selected by i8 in 0..1.
-// v_perm_b32 semantics: sel byte 0..3 picks from src1, 4..7 from src0.
-__device__ __forceinline__ uint32_t ecx_lut(uint32_t t0, uint32_t t1,
-                                            uint32_t t8, uint32_t i7,
-                                            uint32_t i8) {
-  return __builtin_amdgcn_perm(t1, t0, i7) ^ __builtin_amdgcn_perm(0u, t8, i8);
+// GF(2^8) multiply by wave-uniform c over 4 packed bytes, as three
+// v_perm lookups (v_perm sel byte 0..3 picks from src1, 4..7 from src0):
+//   c*x = T7l[x&7] ^ T7h[(x>>4)&7] ^ W[bit3(x) + 2*bit7(x)]
+// where W[s] = c*(8*s0 ^ 128*s1) is a fused 4-entry table covering both
+// high bits in ONE perm (saves a perm + xor per coefficient-dword vs the
+// two 2-entry lookups).
+__device__ __forceinline__ uint32_t ecx_gfmul4(uint32_t l0, uint32_t l1,
+                                               uint32_t h0, uint32_t h1,
+                                               uint32_t w, uint32_t i7l,
+                                               uint32_t i7h, uint32_t i8) {
+  return __builtin_amdgcn_perm(l1, l0, i7l) ^
+         __builtin_amdgcn_perm(h1, h0, i7h) ^
+         __builtin_amdgcn_perm(0u, w, i8);
 }
 
 // VPT = 16-byte vectors per thread per grid-stride iteration. VPT=2 gives
@@ -135,7 +140,7 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
     for (int i = 0; i < n_src; i++) {
       const uint8_t* sp = sbase + (long)s_src[i] * chunk_bytes;
       uint32_t dq[VPT][4];
-      uint32_t i7l[VPT][4], i8l[VPT][4], i7h[VPT][4], i8h[VPT][4];
+      uint32_t i7l[VPT][4], i7h[VPT][4], i8[VPT][4];
 #pragma unroll
       for (int v = 0; v < VPT; v++) {
         v4u d = {0, 0, 0, 0};
@@ -147,9 +152,9 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
 #pragma unroll
         for (int q = 0; q < 4; q++) {
           i7l[v][q] = dq[v][q] & 0x07070707u;
-          i8l[v][q] = (dq[v][q] >> 3) & 0x01010101u;
           i7h[v][q] = (dq[v][q] >> 4) & 0x07070707u;
-          i8h[v][q] = (dq[v][q] >> 7) & 0x01010101u;
+          i8[v][q] = ((dq[v][q] >> 3) & 0x01010101u) |
+                     ((dq[v][q] >> 6) & 0x02020202u);
         }
       }
 #pragma unroll
@@ -165,14 +170,14 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
             for (int q = 0; q < 4; q++) acc[j][v][q] ^= dq[v][q];
         } else {
           const uint32_t* T = &s_tabs[(j * n_src + i) * 6];
-          const uint32_t t0 = T[0], t1 = T[1], t2 = T[2], t3 = T[3],
-                         t4 = T[4], t5 = T[5];
+          const uint32_t l0 = T[0], l1 = T[1], h0 = T[2], h1 = T[3],
+                         w8 = T[4];
 #pragma unroll
           for (int v = 0; v < VPT; v++)
 #pragma unroll
             for (int q = 0; q < 4; q++)
-              acc[j][v][q] ^= ecx_lut(t0, t1, t2, i7l[v][q], i8l[v][q]) ^
-                              ecx_lut(t3, t4, t5, i7h[v][q], i8h[v][q]);
+              acc[j][v][q] ^= ecx_gfmul4(l0, l1, h0, h1, w8, i7l[v][q],
+                                         i7h[v][q], i8[v][q]);
         }
       }
     }
@@ -241,13 +246,13 @@ __global__ __launch_bounds__(256, 2) void ec_gf_slices_kernel(
       const v4u* p4 = reinterpret_cast<const v4u*>(sp + off);
       const v4u d = NT ? __builtin_nontemporal_load(p4) : *p4;
       const uint32_t dq[4] = {d.x, d.y, d.z, d.w};
-      uint32_t i7l[4], i8l[4], i7h[4], i8h[4];
+      uint32_t i7l[4], i7h[4], i8[4];
 #pragma unroll
       for (int q = 0; q < 4; q++) {
         i7l[q] = dq[q] & 0x07070707u;
-        i8l[q] = (dq[q] >> 3) & 0x01010101u;
         i7h[q] = (dq[q] >> 4) & 0x07070707u;
-        i8h[q] = (dq[q] >> 7) & 0x01010101u;
+        i8[q] = ((dq[q] >> 3) & 0x01010101u) |
+                ((dq[q] >> 6) & 0x02020202u);
       }
 #pragma unroll
       for (int j = 0; j < NOUT; j++) {
@@ -258,12 +263,12 @@ __global__ __launch_bounds__(256, 2) void ec_gf_slices_kernel(
           for (int q = 0; q < 4; q++) acc[j][q] ^= dq[q];
         } else {
           const uint32_t* T = &s_tabs[(j * n_src + i) * 6];
-          const uint32_t t0 = T[0], t1 = T[1], t2 = T[2], t3 = T[3],
-                         t4 = T[4], t5 = T[5];
+          const uint32_t l0 = T[0], l1 = T[1], h0 = T[2], h1 = T[3],
+                         w8 = T[4];
 #pragma unroll
           for (int q = 0; q < 4; q++)
-            acc[j][q] ^= ecx_lut(t0, t1, t2, i7l[q], i8l[q]) ^
-                         ecx_lut(t3, t4, t5, i7h[q], i8h[q]);
+            acc[j][q] ^= ecx_gfmul4(l0, l1, h0, h1, w8, i7l[q], i7h[q],
+                                    i8[q]);
         }
       }
     }
@@ -652,19 +657,24 @@ int ecx_dbuf_fill_random(ecx_ctx* ctx, void* dptr, size_t bytes, uint64_t seed,
 
 }  // extern "C"
 
-// Build the 6-dword v_perm tables for one coefficient.
+// Build the 6-dword v_perm tables for one coefficient: T[0..1] = c*x for
+// x in 0..7 (low-3-bit table), T[2..3] = c*(x<<4) for x in 0..7
+// (bits 4-6), T[4] = the fused W table (bit3/bit7 combinations), T[5]
+// unused (kept so the stride stays a friendly 6 dwords).
 static void build_tabs(const ecx::GF8& f, uint8_t c, uint32_t* T) {
-  uint8_t lo[16], hi[16];
+  uint8_t lo[8], hi[8];
   for (int x = 0; x < 8; x++) {
     lo[x] = f.mul(c, (uint8_t)x);
     hi[x] = f.mul(c, (uint8_t)(x << 4));
   }
   T[0] = lo[0] | (lo[1] << 8) | (lo[2] << 16) | ((uint32_t)lo[3] << 24);
   T[1] = lo[4] | (lo[5] << 8) | (lo[6] << 16) | ((uint32_t)lo[7] << 24);
-  T[2] = (uint32_t)f.mul(c, 8) << 8;
-  T[3] = hi[0] | (hi[1] << 8) | (hi[2] << 16) | ((uint32_t)hi[3] << 24);
-  T[4] = hi[4] | (hi[5] << 8) | (hi[6] << 16) | ((uint32_t)hi[7] << 24);
-  T[5] = (uint32_t)f.mul(c, 128) << 8;
+  T[2] = hi[0] | (hi[1] << 8) | (hi[2] << 16) | ((uint32_t)hi[3] << 24);
+  T[3] = hi[4] | (hi[5] << 8) | (hi[6] << 16) | ((uint32_t)hi[7] << 24);
+  uint8_t c8 = f.mul(c, 8), c128 = f.mul(c, 128);
+  T[4] = 0u | ((uint32_t)c8 << 8) | ((uint32_t)c128 << 16) |
+         ((uint32_t)(uint8_t)(c8 ^ c128) << 24);
+  T[5] = 0;
 }
 
 // Populate an EcLaunchParams from a coefficient matrix (n_out x n_src over
