@@ -394,3 +394,22 @@ def test_randomized_config_sweep():
                 assert np.array_equal(chunks[i], full[i]), (trial, i)
         finally:
             ctx.close()
+
+
+@pytest.mark.parametrize("pkt", [256, 512, 2048, 4096])
+def test_cauchy_orig_packetsize_variants(pkt):
+    """packetsize is a live profile key for the bitmatrix technique
+    (ErasureCodeJerasure.h DEFAULT_PACKETSIZE '2048'); every value changes
+    the byte layout and must match the oracle exactly."""
+    k, m = 5, 2
+    C = 8 * pkt * 2
+    rng = np.random.default_rng(pkt)
+    ctx = ceph_amd.EcContext(k, m, "cauchy_orig", device=0, packetsize=pkt)
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        got = ctx.encode_chunks(data)
+        want = oracle.bitmatrix_encode(k, m, data, pkt)
+        for j in range(m):
+            assert np.array_equal(got[j], want[j]), (pkt, j)
+    finally:
+        ctx.close()
