@@ -63,3 +63,27 @@ def test_mi355x_plugin_refuses_cleanly_on_bad_profile():
     r = run_bench("-p", "mi355x", "-P", "technique=no_such_technique",
                   "-P", "k=4", "-P", "m=2", "-s", "65536", "-i", "1")
     assert r.returncode != 0
+
+
+def test_mi355x_reed_sol_r6():
+    """RAID6 technique through the GPU plugin: exhaustive 2-erasure decode
+    with byte verification."""
+    r = run_bench("-p", "mi355x", "-P", "technique=reed_sol_r6_op",
+                  "-P", "k=6", "-P", "m=2", "-s", str(6 * 65536), "-i", "2",
+                  "-w", "decode", "-e", "2", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
+
+
+def test_shec_flags_gpu():
+    r = run_bench("-p", "shec", "-P", "k=4", "-P", "m=3", "-P", "c=2",
+                  "--flags")
+    assert r.returncode == 0, r.stderr
+    assert r.stdout.strip() == ("partialread,partialwrite,zeroinout,"
+                                "paritydelta")
+
+
+def test_clay_flags_gpu():
+    r = run_bench("-p", "clay", "-P", "k=4", "-P", "m=2", "-P", "d=5",
+                  "--flags")
+    assert r.returncode == 0, r.stderr
+    assert r.stdout.strip() == "partialread,requiresubchunks"
