@@ -17,11 +17,13 @@ T_RS_VAN_ISA = 0
 T_CAUCHY_ISA = 1
 T_RS_VAN_JERASURE = 2
 T_CAUCHY_ORIG_JERASURE = 3
+T_RS_VAN_JERASURE_W16 = 4
 TECHNIQUES = {
     "reed_sol_van": T_RS_VAN_ISA,
     "cauchy": T_CAUCHY_ISA,
     "jerasure_reed_sol_van": T_RS_VAN_JERASURE,
     "cauchy_orig": T_CAUCHY_ORIG_JERASURE,
+    "jerasure_reed_sol_van_w16": T_RS_VAN_JERASURE_W16,
 }
 
 _ERR = {
@@ -149,6 +151,8 @@ class EcContext:
     def __init__(self, k, m, technique="reed_sol_van", device=0,
                  n_streams=2, packetsize=2048, w=8):
         t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+        if t == T_RS_VAN_JERASURE_W16:
+            w = 16
         self._h = ctypes.c_void_p()
         self.k, self.m, self.technique = k, m, technique
         self.packetsize = packetsize

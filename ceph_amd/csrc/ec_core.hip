@@ -286,6 +286,141 @@ __global__ __launch_bounds__(256, 2) void ec_gf_slices_kernel(
   }
 }
 
+// ---- GF(2^16) (w=16 jerasure RS-van) path ----
+// Symbols are u16 LE (galois_w16_region_multiply semantics). A coefficient
+// multiply decomposes over the four nibbles of x plus the two fused
+// bit3/bit7-style tables per 8-bit half:
+//   c*x = T0[n0&7] ^ T1[n1&7] ^ T2[n2&7] ^ T3[n3&7]
+//         ^ W01[b3+2*b7] ^ W23[b11+2*b15]
+// Each term is a u16; its lo/hi byte planes are looked up with separate
+// v_perms into separate plane accumulators (selector bytes are replicated
+// per symbol), merged once per output dword with a constant-selector perm.
+// ~24 VALU per coefficient-dword => compute-bound (~1.7-2 TB/s) — w=16 is
+// a compatibility technique, not the fast path.
+struct EcLaunch16 {
+  int n_src;
+  int n_out;
+  int src_ids[ECX_MAX_K];
+  int out_ids[ECX_MAX_OUT];
+  uint8_t cls[ECX_MAX_OUT * ECX_MAX_K];
+  // 20 dwords per (j,i): per plane P in {lo,hi}:
+  //  [T0 pair][T1 pair][T2 pair][T3 pair][W01][W23] = 10 dwords
+  uint32_t tabs[ECX_MAX_OUT * ECX_MAX_K * 20];
+};
+
+#define ECX_PLANE_MERGE_SEL 0x07020500u  // out = [L.b0, H.b1, L.b2, H.b3]
+
+template <int NOUT, bool ACCUM, bool NT>
+__global__ __launch_bounds__(256, 2) void ec_gf16_matmul_kernel(
+    const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
+    const uint8_t* __restrict__ blob, long chunk_bytes,
+    int chunks_per_stripe, long vecs_per_chunk) {
+  const EcLaunch16* pb = (const EcLaunch16*)blob;
+  __shared__ uint32_t s_tabs[ECX_MAX_OUT * ECX_MAX_K * 20];
+  __shared__ int s_src[ECX_MAX_K];
+  __shared__ int s_out[ECX_MAX_OUT];
+  __shared__ uint8_t s_cls[ECX_MAX_OUT * ECX_MAX_K];
+  const int n_src = pb->n_src;
+  for (int t = threadIdx.x; t < NOUT * n_src * 20; t += blockDim.x)
+    s_tabs[t] = pb->tabs[t];
+  for (int t = threadIdx.x; t < n_src; t += blockDim.x)
+    s_src[t] = pb->src_ids[t];
+  for (int t = threadIdx.x; t < NOUT; t += blockDim.x)
+    s_out[t] = pb->out_ids[t];
+  for (int t = threadIdx.x; t < NOUT * n_src; t += blockDim.x)
+    s_cls[t] = pb->cls[t];
+  __syncthreads();
+
+  const long stripe = blockIdx.y;
+  const uint8_t* sbase = buf + stripe * chunks_per_stripe * chunk_bytes;
+  uint8_t* obase = obuf + stripe * chunks_per_stripe * chunk_bytes;
+
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x;
+       p < vecs_per_chunk; p += (long)gridDim.x * blockDim.x) {
+    const long off = p << 4;
+    uint32_t aL[NOUT][4], aH[NOUT][4], aX[NOUT][4];
+#pragma unroll
+    for (int j = 0; j < NOUT; j++)
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+        aL[j][q] = aH[j][q] = 0u;
+        if (ACCUM) {
+          const v4u o = *reinterpret_cast<const v4u*>(
+              obase + (long)s_out[j] * chunk_bytes + off);
+          aX[j][q] = o[q];
+        } else {
+          aX[j][q] = 0u;
+        }
+      }
+
+    for (int i = 0; i < n_src; i++) {
+      const uint8_t* sp = sbase + (long)s_src[i] * chunk_bytes;
+      const v4u* p4 = reinterpret_cast<const v4u*>(sp + off);
+      const v4u d = NT ? __builtin_nontemporal_load(p4) : *p4;
+      const uint32_t dq[4] = {d.x, d.y, d.z, d.w};
+      // selectors (shared across output rows)
+      uint32_t i7[4][4], w01[4], w23[4];
+#pragma unroll
+      for (int q = 0; q < 4; q++) {
+#pragma unroll
+        for (int np = 0; np < 4; np++) {
+          uint32_t nv = (dq[q] >> (4 * np)) & 0x000F000Fu;
+          nv |= nv << 8;
+          i7[np][q] = nv & 0x07070707u;
+        }
+        uint32_t w = ((dq[q] >> 3) & 0x00010001u) |
+                     ((dq[q] >> 6) & 0x00020002u);
+        w01[q] = w | (w << 8);
+        w = ((dq[q] >> 11) & 0x00010001u) | ((dq[q] >> 14) & 0x00020002u);
+        w23[q] = w | (w << 8);
+      }
+#pragma unroll
+      for (int j = 0; j < NOUT; j++) {
+        const int cls = __builtin_amdgcn_readfirstlane(s_cls[j * n_src + i]);
+        if (cls == 0) continue;
+        if (cls == 1) {
+#pragma unroll
+          for (int q = 0; q < 4; q++) aX[j][q] ^= dq[q];
+        } else {
+          const uint32_t* T = &s_tabs[(j * n_src + i) * 20];
+#pragma unroll
+          for (int q = 0; q < 4; q++) {
+            uint32_t l = __builtin_amdgcn_perm(T[1], T[0], i7[0][q]) ^
+                         __builtin_amdgcn_perm(T[3], T[2], i7[1][q]) ^
+                         __builtin_amdgcn_perm(T[5], T[4], i7[2][q]) ^
+                         __builtin_amdgcn_perm(T[7], T[6], i7[3][q]) ^
+                         __builtin_amdgcn_perm(0u, T[8], w01[q]) ^
+                         __builtin_amdgcn_perm(0u, T[9], w23[q]);
+            uint32_t h = __builtin_amdgcn_perm(T[11], T[10], i7[0][q]) ^
+                         __builtin_amdgcn_perm(T[13], T[12], i7[1][q]) ^
+                         __builtin_amdgcn_perm(T[15], T[14], i7[2][q]) ^
+                         __builtin_amdgcn_perm(T[17], T[16], i7[3][q]) ^
+                         __builtin_amdgcn_perm(0u, T[18], w01[q]) ^
+                         __builtin_amdgcn_perm(0u, T[19], w23[q]);
+            aL[j][q] ^= l;
+            aH[j][q] ^= h;
+          }
+        }
+      }
+    }
+
+#pragma unroll
+    for (int j = 0; j < NOUT; j++) {
+      v4u o;
+#pragma unroll
+      for (int q = 0; q < 4; q++)
+        o[q] = aX[j][q] ^ __builtin_amdgcn_perm(aH[j][q], aL[j][q],
+                                                ECX_PLANE_MERGE_SEL);
+      v4u* dp = reinterpret_cast<v4u*>(obase + (long)s_out[j] * chunk_bytes +
+                                       off);
+      if (NT)
+        __builtin_nontemporal_store(o, dp);
+      else
+        *dp = o;
+    }
+  }
+}
+
 // ---- jerasure bitmatrix (Cauchy-original) path ----
 // Packet-sliced XOR gather: each chunk is a stream of superwords (w packets
 // of `pkt` bytes); coding packet row r = XOR of the data packets its bit
@@ -435,13 +570,20 @@ struct BitPlan {
   std::vector<uint8_t> rows;  // (n_erased*w) x (k*w) bits
 };
 
+struct Decode16Plan {
+  std::vector<int> survivors;
+  std::vector<int> erased;
+  std::vector<uint16_t> rows;  // n_erased x k u16
+};
+
 }  // namespace
 
 struct ecx_ctx {
   int k = 0, m = 0, technique = 0, device = 0;
   int w = 8, pkt = 2048;  // bitmatrix techniques only (jerasure packetsize)
-  std::vector<uint8_t> gen;     // (k+m) x k
-  std::vector<uint8_t> bitmat;  // (m*w) x (k*w) bits (bitmatrix techniques)
+  std::vector<uint8_t> gen;      // (k+m) x k (w=8 techniques)
+  std::vector<uint16_t> gen16;   // (k+m) x k (w=16 technique)
+  std::vector<uint8_t> bitmat;   // (m*w) x (k*w) bits (bitmatrix techniques)
   std::vector<Slot> slots;
   // decode-plan LRU keyed by present_mask (exact signature for fixed
   // (k,m,technique) — the analogue of ErasureCodeIsaTableCache's
@@ -451,6 +593,7 @@ struct ecx_ctx {
   std::map<uint64_t, std::pair<DecodePlan, std::list<uint64_t>::iterator>> lru;
   std::list<uint64_t> lru_order;
   std::map<uint64_t, BitPlan> bit_lru;  // bitmatrix decode plans
+  std::map<uint64_t, Decode16Plan> lru16;  // w=16 decode plans
   static constexpr size_t LRU_DEPTH = 4096;
   // host-pointer calls round-robin the stream slots so concurrent plugin
   // threads (the OSD's PG workers) overlap instead of serialising on one
@@ -458,6 +601,7 @@ struct ecx_ctx {
   std::atomic<unsigned> rr{0};
 
   bool is_bitmatrix() const { return technique == ECX_T_CAUCHY_ORIG_JERASURE; }
+  bool is_w16() const { return technique == ECX_T_RS_VAN_JERASURE_W16; }
 };
 
 static int map_hip(hipError_t e) {
@@ -485,8 +629,9 @@ int ecx_device_count(void) {
 
 int ecx_create2(int k, int m, int technique, int w, int packetsize,
                 int device, int n_streams, ecx_ctx** out) {
+  const bool want16 = (technique == ECX_T_RS_VAN_JERASURE_W16);
   if (!out || k < 2 || m < 1 || k > ECX_MAX_K || m > ECX_MAX_K ||
-      n_streams < 1 || n_streams > 64 || w != 8)
+      n_streams < 1 || n_streams > 64 || w != (want16 ? 16 : 8))
     return ECX_ERR_INVAL;
   if (technique == ECX_T_CAUCHY_ORIG_JERASURE &&
       (packetsize < 16 || packetsize % 16 || k > 16))
@@ -500,7 +645,12 @@ int ecx_create2(int k, int m, int technique, int w, int packetsize,
   ctx->device = device;
   ctx->w = w;
   ctx->pkt = packetsize;
-  if (!ecx::gen_matrix(technique, ctx->gen, k, m)) {
+  if (want16) {
+    if (!ecx::gen_matrix_rs_van_jerasure_w16(ctx->gen16, k, m)) {
+      delete ctx;
+      return ECX_ERR_INVAL;
+    }
+  } else if (!ecx::gen_matrix(technique, ctx->gen, k, m)) {
     delete ctx;
     return ECX_ERR_INVAL;
   }
@@ -557,8 +707,14 @@ int ecx_k(const ecx_ctx* ctx) { return ctx ? ctx->k : ECX_ERR_INVAL; }
 int ecx_m(const ecx_ctx* ctx) { return ctx ? ctx->m : ECX_ERR_INVAL; }
 
 int ecx_get_matrix(const ecx_ctx* ctx, uint8_t* out) {
-  if (!ctx || !out) return ECX_ERR_INVAL;
+  if (!ctx || !out || ctx->is_w16()) return ECX_ERR_INVAL;
   std::memcpy(out, ctx->gen.data(), ctx->gen.size());
+  return ECX_OK;
+}
+
+int ecx_get_matrix16(const ecx_ctx* ctx, uint16_t* out) {
+  if (!ctx || !out || !ctx->is_w16()) return ECX_ERR_INVAL;
+  std::memcpy(out, ctx->gen16.data(), ctx->gen16.size() * 2);
   return ECX_OK;
 }
 
@@ -573,9 +729,9 @@ unsigned ecx_chunk_size(const ecx_ctx* ctx, unsigned stripe_width) {
     unsigned padded = stripe_width + (tail ? align - tail : 0);
     return padded / ctx->k;
   }
-  if (ctx->technique == ECX_T_RS_VAN_JERASURE) {
-    // ErasureCodeJerasure.cc:85-108, w=8, per_chunk_alignment=false
-    unsigned align = (unsigned)ctx->k * 8u * 4u;
+  if (ctx->technique == ECX_T_RS_VAN_JERASURE || ctx->is_w16()) {
+    // ErasureCodeJerasure.cc:85-108, per_chunk_alignment=false
+    unsigned align = (unsigned)ctx->k * ctx->w * 4u;
     unsigned tail = stripe_width % align;
     unsigned padded = stripe_width + (tail ? align - tail : 0);
     return padded / ctx->k;
@@ -1042,6 +1198,140 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   return ECX_OK;
 }
 
+// w=16 tables: 20 dwords per coefficient (see EcLaunch16).
+static void build_tabs16(const ecx::GF16& f, uint16_t c, uint32_t* T) {
+  for (int plane = 0; plane < 2; plane++) {
+    uint32_t* P = T + plane * 10;
+    for (int np = 0; np < 4; np++) {
+      uint8_t e[8];
+      for (int v = 0; v < 8; v++) {
+        uint16_t r = f.mul(c, (uint16_t)(v << (4 * np)));
+        e[v] = plane ? (uint8_t)(r >> 8) : (uint8_t)r;
+      }
+      P[np * 2] = e[0] | (e[1] << 8) | (e[2] << 16) | ((uint32_t)e[3] << 24);
+      P[np * 2 + 1] =
+          e[4] | (e[5] << 8) | (e[6] << 16) | ((uint32_t)e[7] << 24);
+    }
+    uint16_t w01[4] = {0, f.mul(c, 8), f.mul(c, 128),
+                       (uint16_t)(f.mul(c, 8) ^ f.mul(c, 128))};
+    uint16_t w23[4] = {0, f.mul(c, (uint16_t)(1u << 11)),
+                       f.mul(c, (uint16_t)(1u << 15)),
+                       (uint16_t)(f.mul(c, (uint16_t)(1u << 11)) ^
+                                  f.mul(c, (uint16_t)(1u << 15)))};
+    uint32_t a = 0, b = 0;
+    for (int sidx = 0; sidx < 4; sidx++) {
+      a |= (uint32_t)(plane ? (w01[sidx] >> 8) : (w01[sidx] & 0xff))
+           << (8 * sidx);
+      b |= (uint32_t)(plane ? (w23[sidx] >> 8) : (w23[sidx] & 0xff))
+           << (8 * sidx);
+    }
+    P[8] = a;
+    P[9] = b;
+  }
+}
+
+// Launch the w=16 kernel for <= 4 output rows per group.
+static int run_matmul16(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
+                        uint8_t* d_obuf, const int* src_ids, int n_src,
+                        const int* out_ids, int n_out,
+                        const uint16_t* coeff, long n_stripes,
+                        size_t chunk_bytes, bool accum, bool time_all) {
+  if (!ctx || slot_i < 0 || slot_i >= (int)ctx->slots.size() || n_src < 1 ||
+      n_src > ECX_MAX_K || n_out < 1)
+    return ECX_ERR_INVAL;
+  if (chunk_bytes % 16 || n_stripes <= 0 || n_stripes > 65535)
+    return ECX_ERR_INVAL;
+  Slot& s = ctx->slots[slot_i];
+  std::lock_guard<std::recursive_mutex> g(s.mu);
+  HIP_TRY(hipSetDevice(ctx->device));
+  static const int env_nt = [] {
+    const char* v = getenv("ECX_NT");
+    return v ? atoi(v) : 1;
+  }();
+  const ecx::GF16& f = ecx::gf16();
+  const long vecs = (long)(chunk_bytes >> 4);
+  int gx = (int)std::min<long>((vecs + 256 * 2 - 1) / (256 * 2), 1024);
+  if (gx < 1) gx = 1;
+  long tiles = (vecs + 255) / 256;
+  while ((long)gx * n_stripes < 2048 && gx < tiles) gx *= 2;
+  dim3 grid(gx, (unsigned)n_stripes);
+  const int cps = ctx->k + ctx->m;
+
+  for (int j0 = 0; j0 < n_out; j0 += 4) {
+    int nj = std::min(4, n_out - j0);
+    EcLaunch16 p;
+    std::memset(&p, 0, sizeof(p));
+    p.n_src = n_src;
+    p.n_out = nj;
+    for (int i = 0; i < n_src; i++) p.src_ids[i] = src_ids[i];
+    for (int j = 0; j < nj; j++) p.out_ids[j] = out_ids[j0 + j];
+    for (int j = 0; j < nj; j++)
+      for (int i = 0; i < n_src; i++) {
+        uint16_t c = coeff[(size_t)(j0 + j) * n_src + i];
+        uint8_t cls = (c == 0) ? 0 : (c == 1 ? 1 : 2);
+        p.cls[j * n_src + i] = cls;
+        if (cls == 2) build_tabs16(f, c, &p.tabs[(j * n_src + i) * 20]);
+      }
+    size_t blob = sizeof(EcLaunch16);
+    int r = ensure_jobs(ctx, s, blob);
+    if (r != ECX_OK) return r;
+    HIP_TRY(hipEventSynchronize(s.ev_jobs));
+    std::memcpy(s.h_jobs, &p, blob);
+    HIP_TRY(hipMemcpyAsync(s.d_jobs, s.h_jobs, blob, hipMemcpyHostToDevice,
+                           s.stream));
+    HIP_TRY(hipEventRecord(s.ev_jobs, s.stream));
+    if (j0 == 0 && time_all) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
+#define ECX_L16(NO, AC, NTF)                                                \
+  hipLaunchKernelGGL((ec_gf16_matmul_kernel<NO, AC, NTF>), grid, dim3(256), \
+                     0, s.stream, d_buf, d_obuf, s.d_jobs,                  \
+                     (long)chunk_bytes, cps, vecs)
+#define ECX_D16(NO)                                \
+  case NO:                                         \
+    if (accum) {                                   \
+      if (env_nt) ECX_L16(NO, true, true);         \
+      else ECX_L16(NO, true, false);               \
+    } else {                                       \
+      if (env_nt) ECX_L16(NO, false, true);        \
+      else ECX_L16(NO, false, false);              \
+    }                                              \
+    break;
+    switch (nj) {
+      ECX_D16(1)
+      ECX_D16(2)
+      ECX_D16(3)
+      ECX_D16(4)
+      default:
+        return ECX_ERR_INVAL;
+    }
+#undef ECX_D16
+#undef ECX_L16
+    HIP_TRY(hipGetLastError());
+    if (time_all) {
+      HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
+      s.timed = true;
+    }
+  }
+  return ECX_OK;
+}
+
+static int get_plan16(ecx_ctx* ctx, uint64_t present_mask,
+                      Decode16Plan& out) {
+  std::lock_guard<std::mutex> g(ctx->lru_mu);
+  auto it = ctx->lru16.find(present_mask);
+  if (it != ctx->lru16.end()) {
+    out = it->second;
+    return ECX_OK;
+  }
+  Decode16Plan plan;
+  if (!ecx::compose_decode_rows16(ctx->gen16, ctx->k, ctx->m, present_mask,
+                                  plan.survivors, plan.erased, plan.rows))
+    return ECX_ERR_IO;
+  if (ctx->lru16.size() > ecx_ctx::LRU_DEPTH) ctx->lru16.clear();
+  ctx->lru16.emplace(present_mask, plan);
+  out = plan;
+  return ECX_OK;
+}
+
 static int get_bit_plan(ecx_ctx* ctx, uint64_t present_mask, BitPlan& out) {
   std::lock_guard<std::mutex> g(ctx->lru_mu);
   auto it = ctx->bit_lru.find(present_mask);
@@ -1101,6 +1391,11 @@ int ecx_encode_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
                          src_ids, k, out_ids, m, ctx->bitmat.data(),
                          n_stripes, chunk_bytes, true);
   }
+  if (ctx->is_w16())
+    return run_matmul16(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr,
+                        src_ids, k, out_ids, m,
+                        ctx->gen16.data() + (size_t)k * k, n_stripes,
+                        chunk_bytes, false, true);
   const uint8_t* rows = ctx->gen.data() + (size_t)k * k;
   return run_matmul(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr, src_ids,
                     k, out_ids, m, rows, nullptr, n_stripes, chunk_bytes,
@@ -1120,6 +1415,16 @@ int ecx_decode_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
                          plan.survivors.data(), ctx->k, plan.erased.data(),
                          (int)plan.erased.size(), plan.rows.data(),
                          n_stripes, chunk_bytes, false);
+  }
+  if (ctx->is_w16()) {
+    Decode16Plan plan;
+    int r = get_plan16(ctx, present_mask, plan);
+    if (r != ECX_OK) return r;
+    if (plan.erased.empty()) return ECX_OK;
+    return run_matmul16(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr,
+                        plan.survivors.data(), ctx->k, plan.erased.data(),
+                        (int)plan.erased.size(), plan.rows.data(),
+                        n_stripes, chunk_bytes, false, false);
   }
   DecodePlan plan;
   int r = get_decode_plan(ctx, present_mask, plan);
@@ -1154,6 +1459,15 @@ int ecx_apply_delta_dev(ecx_ctx* ctx, const void* d_delta, int data_shard,
   if (!ctx || data_shard < 0 || data_shard >= ctx->k || coding_shard < ctx->k ||
       coding_shard >= ctx->k + ctx->m)
     return ECX_ERR_INVAL;
+  if (ctx->is_w16()) {
+    uint16_t c16 = ctx->gen16[(size_t)coding_shard * ctx->k + data_shard];
+    int sids[1] = {0}, oids[1] = {0};
+    uint16_t cf[1] = {c16};
+    return run_matmul16(ctx, slot, (const uint8_t*)d_delta,
+                        (uint8_t*)d_parity, sids, 1, oids, 1, cf, 1, bytes,
+                        true, false);
+  }
+  if (ctx->is_bitmatrix()) return ECX_ERR_INVAL;  // schedule deltas: later
   uint8_t c = ctx->gen[(size_t)coding_shard * ctx->k + data_shard];
   int src_ids[1] = {0};
   int out_ids[1] = {0};
@@ -1166,7 +1480,8 @@ int ecx_apply_delta_dev(ecx_ctx* ctx, const void* d_delta, int data_shard,
 
 int ecx_encode_slices(ecx_ctx* ctx, void* const* d_chunks,
                       const size_t* bytes, int n_slices, int slot) {
-  if (!ctx || !d_chunks || !bytes) return ECX_ERR_INVAL;
+  if (!ctx || !d_chunks || !bytes || ctx->is_bitmatrix() || ctx->is_w16())
+    return ECX_ERR_INVAL;
   int k = ctx->k, m = ctx->m;
   int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
   for (int i = 0; i < k; i++) src_ids[i] = i;
@@ -1178,7 +1493,8 @@ int ecx_encode_slices(ecx_ctx* ctx, void* const* d_chunks,
 int ecx_decode_slices(ecx_ctx* ctx, void* const* d_chunks,
                       const size_t* bytes, int n_slices,
                       uint64_t present_mask, int slot) {
-  if (!ctx || !d_chunks || !bytes) return ECX_ERR_INVAL;
+  if (!ctx || !d_chunks || !bytes || ctx->is_bitmatrix() || ctx->is_w16())
+    return ECX_ERR_INVAL;
   DecodePlan plan;
   int r = get_decode_plan(ctx, present_mask, plan);
   if (r != ECX_OK) return r;
@@ -1192,13 +1508,15 @@ int ecx_decode_slices(ecx_ctx* ctx, void* const* d_chunks,
 static int ensure_stage(ecx_ctx* ctx, Slot& s, size_t bytes);
 
 int ecx_set_matrix(ecx_ctx* ctx, const uint8_t* coding_rows) {
-  if (!ctx || !coding_rows || ctx->is_bitmatrix()) return ECX_ERR_INVAL;
+  if (!ctx || !coding_rows || ctx->is_bitmatrix() || ctx->is_w16())
+    return ECX_ERR_INVAL;
   std::lock_guard<std::mutex> g(ctx->lru_mu);
   std::memcpy(ctx->gen.data() + (size_t)ctx->k * ctx->k, coding_rows,
               (size_t)ctx->m * ctx->k);
   ctx->lru.clear();
   ctx->lru_order.clear();
   ctx->bit_lru.clear();
+  ctx->lru16.clear();
   return ECX_OK;
 }
 
@@ -1315,7 +1633,17 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
   int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
   for (int i = 0; i < k; i++) src_ids[i] = i;
   for (int j = 0; j < m; j++) out_ids[j] = k + j;
-  if (ctx->is_bitmatrix()) {
+  if (ctx->is_w16()) {
+    // per-slice NULL handled via cls in run_matmul16? no: zero-fill stage
+    for (int i = 0; i < k; i++)
+      if (src_null[i])
+        HIP_TRY(hipMemsetAsync(s.d_stage + (size_t)i * chunk_bytes, 0,
+                               chunk_bytes, s.stream));
+    int rr = run_matmul16(ctx, si, s.d_stage, s.d_stage, src_ids, k,
+                          out_ids, m, ctx->gen16.data() + (size_t)k * k, 1,
+                          chunk_bytes, false, false);
+    if (rr != ECX_OK) return rr;
+  } else if (ctx->is_bitmatrix()) {
     // zeros-chunk convention: materialise zeros in the stage buffer
     for (int i = 0; i < k; i++)
       if (src_null[i])
@@ -1357,6 +1685,41 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
                            uint64_t present_mask, size_t chunk_bytes) {
   if (!ctx || !chunks || chunk_bytes % 16) return ECX_ERR_INVAL;
   int k = ctx->k, m = ctx->m, n = k + m;
+
+  if (ctx->is_w16()) {
+    Decode16Plan plan;
+    int r = get_plan16(ctx, present_mask, plan);
+    if (r != ECX_OK) return r;
+    if (plan.erased.empty()) return ECX_OK;
+    const int si = (int)(ctx->rr++ % ctx->slots.size());
+    Slot& s = ctx->slots[si];
+    std::lock_guard<std::recursive_mutex> g(s.mu);
+    r = ensure_stage(ctx, s, (size_t)n * chunk_bytes);
+    if (r != ECX_OK) return r;
+    HIP_TRY(hipSetDevice(ctx->device));
+    for (int i = 0; i < k; i++) {
+      int id = plan.survivors[i];
+      if (!chunks[id]) {
+        HIP_TRY(hipMemsetAsync(s.d_stage + (size_t)id * chunk_bytes, 0,
+                               chunk_bytes, s.stream));
+        continue;
+      }
+      HIP_TRY(hipMemcpyAsync(s.d_stage + (size_t)id * chunk_bytes,
+                             chunks[id], chunk_bytes, hipMemcpyHostToDevice,
+                             s.stream));
+    }
+    r = run_matmul16(ctx, si, s.d_stage, s.d_stage, plan.survivors.data(),
+                     k, plan.erased.data(), (int)plan.erased.size(),
+                     plan.rows.data(), 1, chunk_bytes, false, false);
+    if (r != ECX_OK) return r;
+    for (int e : plan.erased) {
+      if (!chunks[e]) return ECX_ERR_INVAL;
+      HIP_TRY(hipMemcpyAsync(chunks[e], s.d_stage + (size_t)e * chunk_bytes,
+                             chunk_bytes, hipMemcpyDeviceToHost, s.stream));
+    }
+    HIP_TRY(hipStreamSynchronize(s.stream));
+    return ECX_OK;
+  }
 
   if (ctx->is_bitmatrix()) {
     BitPlan plan;

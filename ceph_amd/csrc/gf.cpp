@@ -299,6 +299,143 @@ bool compose_decode_rows(const std::vector<uint8_t> &gen, int k, int m,
   return true;
 }
 
+// ---- GF(2^16) (w=16 jerasure RS-van) ----
+
+GF16::GF16() : log(65536), exp(65536) {
+  unsigned v = 1;
+  for (int i = 0; i < 65535; i++) {
+    exp[i] = (uint16_t)v;
+    log[v] = (uint16_t)i;
+    v <<= 1;
+    if (v & 0x10000) v ^= 0x1100B;
+  }
+  exp[65535] = exp[0];
+  log[0] = 0;
+}
+
+const GF16 &gf16() {
+  static const GF16 f;
+  return f;
+}
+
+bool gen_matrix_rs_van_jerasure_w16(std::vector<uint16_t> &a, int k, int m) {
+  const GF16 &f = gf16();
+  int rows = k + m, cols = k;
+  if (k < 1 || m < 0 || rows > 65535) return false;
+  a.assign((size_t)rows * cols, 0);
+  a[0] = 1;
+  if (rows > 1) a[(size_t)(rows - 1) * cols + (cols - 1)] = 1;
+  for (int i = 1; i < rows - 1; i++) {
+    uint16_t v = 1;
+    for (int j = 0; j < cols; j++) {
+      a[(size_t)i * cols + j] = v;
+      v = f.mul(v, (uint16_t)i);
+    }
+  }
+  for (int i = 1; i < cols; i++) {
+    int j = i;
+    while (j < rows && a[(size_t)j * cols + i] == 0) j++;
+    if (j >= rows) return false;
+    if (j != i)
+      for (int c = 0; c < cols; c++)
+        std::swap(a[(size_t)j * cols + c], a[(size_t)i * cols + c]);
+    uint16_t piv = a[(size_t)i * cols + i];
+    if (piv != 1) {
+      uint16_t inv = f.div(1, piv);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + i] = f.mul(inv, a[(size_t)r * cols + i]);
+    }
+    for (int c = 0; c < cols; c++) {
+      uint16_t t = a[(size_t)i * cols + c];
+      if (c != i && t != 0)
+        for (int r = 0; r < rows; r++)
+          a[(size_t)r * cols + c] ^= f.mul(t, a[(size_t)r * cols + i]);
+    }
+  }
+  for (int j = 0; j < cols; j++) {
+    uint16_t t = a[(size_t)cols * cols + j];
+    if (t != 0 && t != 1) {
+      uint16_t inv = f.div(1, t);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + j] = f.mul(inv, a[(size_t)r * cols + j]);
+      for (int c = 0; c < cols; c++)
+        a[(size_t)j * cols + c] = f.mul(t, a[(size_t)j * cols + c]);
+    }
+  }
+  return true;
+}
+
+bool gf16_invert(const uint16_t *in, uint16_t *out, int k) {
+  const GF16 &f = gf16();
+  std::vector<uint16_t> w(in, in + (size_t)k * k);
+  std::memset(out, 0, (size_t)k * k * 2);
+  for (int i = 0; i < k; i++) out[(size_t)k * i + i] = 1;
+  for (int i = 0; i < k; i++) {
+    if (!w[(size_t)k * i + i]) {
+      int j = i + 1;
+      while (j < k && !w[(size_t)k * j + i]) j++;
+      if (j >= k) return false;
+      for (int c = 0; c < k; c++) {
+        std::swap(w[(size_t)k * i + c], w[(size_t)k * j + c]);
+        std::swap(out[(size_t)k * i + c], out[(size_t)k * j + c]);
+      }
+    }
+    uint16_t inv = f.inv(w[(size_t)k * i + i]);
+    for (int c = 0; c < k; c++) {
+      w[(size_t)k * i + c] = f.mul(inv, w[(size_t)k * i + c]);
+      out[(size_t)k * i + c] = f.mul(inv, out[(size_t)k * i + c]);
+    }
+    for (int r = 0; r < k; r++) {
+      if (r == i) continue;
+      uint16_t t = w[(size_t)k * r + i];
+      if (!t) continue;
+      for (int c = 0; c < k; c++) {
+        w[(size_t)k * r + c] ^= f.mul(t, w[(size_t)k * i + c]);
+        out[(size_t)k * r + c] ^= f.mul(t, out[(size_t)k * i + c]);
+      }
+    }
+  }
+  return true;
+}
+
+bool compose_decode_rows16(const std::vector<uint16_t> &gen, int k, int m,
+                           uint64_t present_mask,
+                           std::vector<int> &survivors,
+                           std::vector<int> &erased,
+                           std::vector<uint16_t> &rows) {
+  const GF16 &f = gf16();
+  int n = k + m;
+  survivors.clear();
+  erased.clear();
+  for (int i = 0; i < n; i++) {
+    if (present_mask & (1ull << i)) {
+      if ((int)survivors.size() < k) survivors.push_back(i);
+    } else {
+      erased.push_back(i);
+    }
+  }
+  if ((int)survivors.size() < k || (int)erased.size() > m) return false;
+  std::vector<uint16_t> b((size_t)k * k), d((size_t)k * k);
+  for (int i = 0; i < k; i++)
+    std::memcpy(&b[(size_t)i * k], &gen[(size_t)survivors[i] * k], k * 2);
+  if (!gf16_invert(b.data(), d.data(), k)) return false;
+  rows.assign(erased.size() * (size_t)k, 0);
+  for (size_t p = 0; p < erased.size(); p++) {
+    int e = erased[p];
+    if (e < k) {
+      std::memcpy(&rows[p * k], &d[(size_t)e * k], k * 2);
+    } else {
+      for (int i = 0; i < k; i++) {
+        uint16_t s = 0;
+        for (int j = 0; j < k; j++)
+          s ^= f.mul(d[(size_t)j * k + i], gen[(size_t)e * k + j]);
+        rows[p * k + i] = s;
+      }
+    }
+  }
+  return true;
+}
+
 // ---- SHEC (restated from the reference's in-tree implementation) ----
 
 // shec_calc_recovery_efficiency1 (ErasureCodeShec.cc:660-697)

@@ -74,6 +74,37 @@ bool compose_decode_rows(const std::vector<uint8_t> &gen, int k, int m,
                          std::vector<int> &erased,
                          std::vector<uint8_t> &rows);
 
+// GF(2^16) (w=16 jerasure reed_sol_van): gf-complete default w=16 field
+// (poly 0x1100B) + the same big-Vandermonde construction with u16
+// coefficients/symbols (galois_w16_region_multiply semantics,
+// ErasureCodeJerasure.cc:316-319).
+struct GF16 {
+  std::vector<uint16_t> log, exp;
+  GF16();
+  uint16_t mul(uint16_t a, uint16_t b) const {
+    if (!a || !b) return 0;
+    unsigned s = log[a] + log[b];
+    if (s >= 65535) s -= 65535;
+    return exp[s];
+  }
+  uint16_t inv(uint16_t a) const { return a ? exp[65535 - log[a]] : 0; }
+  uint16_t div(uint16_t a, uint16_t b) const {
+    if (!a || !b) return 0;
+    int s = (int)log[a] - (int)log[b];
+    if (s < 0) s += 65535;
+    return exp[s];
+  }
+};
+const GF16 &gf16();
+// full (k+m) x k generator, identity top
+bool gen_matrix_rs_van_jerasure_w16(std::vector<uint16_t> &a, int k, int m);
+bool gf16_invert(const uint16_t *in, uint16_t *out, int k);
+bool compose_decode_rows16(const std::vector<uint16_t> &gen, int k, int m,
+                           uint64_t present_mask,
+                           std::vector<int> &survivors,
+                           std::vector<int> &erased,
+                           std::vector<uint16_t> &rows);
+
 // SHEC (shingled EC) matrix + decode search, restated from the reference's
 // OWN in-tree implementation (src/erasure-code/shec/ErasureCodeShec.cc —
 // unlike the GF submodules this algorithm is fully present):

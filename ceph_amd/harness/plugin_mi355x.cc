@@ -73,8 +73,14 @@ class ErasureCodeMi355x final : public ErasureCode {
     if (is_bitmatrix())
       err |= to_int("packetsize", profile, &packetsize_, "2048", ss);
     err |= sanity_check_k_m(k_, m_, ss);
-    if (w_ != 8) {
-      if (ss) *ss << "mi355x: w=" << w_ << " must be 8\n";
+    // w=16 is supported for the jerasure RS-van technique (GF(2^16)
+    // matrix, ErasureCodeJerasure.cc:421 accepts w in {8,16,32}; w=32 is
+    // not implemented here)
+    if (!(w_ == 8 ||
+          (w_ == 16 && technique_ == "jerasure_reed_sol_van"))) {
+      if (ss)
+        *ss << "mi355x: w=" << w_
+            << " must be 8 (or 16 with jerasure_reed_sol_van)\n";
       err = -EINVAL;
     }
     if (technique_id(technique_) < 0) {
@@ -93,8 +99,11 @@ class ErasureCodeMi355x final : public ErasureCode {
   int init(ErasureCodeProfile &profile, std::ostream *ss) override {
     int err = parse(profile, ss);
     if (err) return err;
-    int r = ecx_create2(k_, m_, technique_id(technique_), w_, packetsize_,
-                        device_, streams_, &ctx_);
+    int tid = technique_id(technique_);
+    if (w_ == 16 && technique_ == "jerasure_reed_sol_van")
+      tid = ECX_T_RS_VAN_JERASURE_W16;
+    int r = ecx_create2(k_, m_, tid, w_, packetsize_, device_, streams_,
+                        &ctx_);
     if (r != ECX_OK) {
       if (ss)
         *ss << "mi355x: ecx_create failed (" << r

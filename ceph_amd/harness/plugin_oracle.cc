@@ -30,6 +30,7 @@ int technique_id(const std::string &t) {
 
 class ErasureCodeOracle final : public ErasureCode {
   int k_ = 0, m_ = 0, packetsize_ = 2048, w_ = 8;
+  std::vector<uint16_t> gen16_;
   std::string technique_;
   std::vector<uint8_t> gen_;
   std::vector<uint8_t> bitmat_;
@@ -53,7 +54,21 @@ class ErasureCodeOracle final : public ErasureCode {
     int err = ErasureCode::parse(profile, ss);
     err |= to_int("k", profile, &k_, "8", ss);
     err |= to_int("m", profile, &m_, "3", ss);
+    err |= to_int("w", profile, &w_, "8", ss);
     err |= sanity_check_k_m(k_, m_, ss);
+    if (w_ == 16 && technique_ == "jerasure_reed_sol_van") {
+      if (err) return err;
+      profile["technique"] = technique_;
+      ecref_gf16_init();
+      gen16_.resize((size_t)(k_ + m_) * k_);
+      if (ecref_matrix_rs_vandermonde_jerasure_w16(gen16_.data(), k_, m_))
+        return -EINVAL;
+      return ErasureCode::init(profile, ss);
+    }
+    if (w_ != 8) {
+      if (ss) *ss << "oracle: w=" << w_ << " unsupported here\n";
+      return -EINVAL;
+    }
     int t = technique_id(technique_);
     if (t < 0) {
       if (ss) *ss << "oracle: unknown technique " << technique_ << "\n";
@@ -85,7 +100,10 @@ class ErasureCodeOracle final : public ErasureCode {
     return ErasureCode::init(profile, ss);
   }
 
+  bool is_w16() const { return !gen16_.empty(); }
+
   unsigned int get_chunk_size(unsigned int stripe_width) const override {
+    if (is_w16()) return ecref_chunk_size_jerasure(k_, 16, stripe_width);
     if (is_bitmatrix()) {
       // ErasureCodeJerasureCauchy::get_alignment rule
       unsigned align = (unsigned)k_ * w_ * packetsize_ * 4u;
@@ -126,6 +144,11 @@ class ErasureCodeOracle final : public ErasureCode {
       }
     }
     if (!size) return 0;
+    if (is_w16()) {
+      ecref_encode16(k_, m_, gen16_.data() + (size_t)k_ * k_, data, parity,
+                     size);
+      return 0;
+    }
     if (is_bitmatrix())
       return ecref_bitmatrix_encode(k_, m_, w_, bitmat_.data(), data, parity,
                                     size, packetsize_) == 0 ? 0 : -EINVAL;
@@ -158,6 +181,8 @@ class ErasureCodeOracle final : public ErasureCode {
         chunks[i] = scratch.back().c_str();
       }
     }
+    if (is_w16())
+      return ecref_decode16(chunks, present, k_, m_, size) == 0 ? 0 : -EIO;
     if (is_bitmatrix())
       return ecref_bitmatrix_decode(k_, m_, w_, bitmat_.data(), chunks,
                                     present, size, packetsize_) == 0

@@ -64,6 +64,11 @@ enum ecx_technique {
    * chunk sizes must be multiples of w*packetsize
    * (ErasureCodeJerasureCauchy::get_alignment, ErasureCodeJerasure.cc:522-536) */
   ECX_T_CAUCHY_ORIG_JERASURE = 3,
+  /* jerasure reed_sol_van with w=16: GF(2^16) over gf-complete's 0x1100B,
+   * u16 LE symbols (galois_w16_region_multiply semantics). Select via
+   * ecx_create2(..., w=16, ...) with this id. Compute-bound compatibility
+   * technique; chunk sizes are multiples of k*w*4 / k = 64 B. */
+  ECX_T_RS_VAN_JERASURE_W16 = 4,
 };
 
 enum ecx_err {
@@ -103,6 +108,9 @@ ECX_API int ecx_m(const ecx_ctx *ctx);
 /* Generator matrix readback for tests/verification: fills (k+m)*k bytes
  * (identity top, coding rows below), isa-l row-major layout. */
 ECX_API int ecx_get_matrix(const ecx_ctx *ctx, uint8_t *out);
+/* w=16 variant: fills (k+m)*k u16 entries; EINVAL on w=8 contexts (and
+ * vice versa for ecx_get_matrix). */
+ECX_API int ecx_get_matrix16(const ecx_ctx *ctx, uint16_t *out);
 
 /* Chunk-size rule of the technique (a7 in SURVEY §8):
  * ECX_T_*_ISA: ceil(width/k) rounded up to 32 (ErasureCodeIsa.cc:65-79);

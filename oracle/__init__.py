@@ -64,6 +64,8 @@ _ref.ecref_chunk_size_jerasure.argtypes = [ctypes.c_int, ctypes.c_int,
 _ref.ecref_matrix_cauchy_orig_jerasure.restype = ctypes.c_int
 _ref.ecref_bitmatrix_encode.restype = ctypes.c_int
 _ref.ecref_bitmatrix_decode.restype = ctypes.c_int
+_ref.ecref_matrix_rs_vandermonde_jerasure_w16.restype = ctypes.c_int
+_ref.ecref_decode16.restype = ctypes.c_int
 _cpu.eccpu_encode_batch.restype = ctypes.c_int
 _cpu.eccpu_decode_batch.restype = ctypes.c_int
 _cpu.eccpu_threads.restype = ctypes.c_int
@@ -199,6 +201,45 @@ def bitmatrix_decode(k, m, chunks, present, packetsize, w=8):
         ctypes.c_size_t(chunks[0].nbytes), packetsize)
     if r != 0:
         raise ValueError(f"bitmatrix_decode failed: {r}")
+    return chunks
+
+
+def gf16_mul(a, b):
+    _ref.ecref_gf16_init()
+    _ref.ecref_gf16_mul.restype = ctypes.c_uint16
+    return _ref.ecref_gf16_mul(ctypes.c_uint16(a), ctypes.c_uint16(b))
+
+
+def matrix_w16(k, m):
+    _ref.ecref_gf16_init()
+    a = np.zeros((k + m, k), dtype=np.uint16)
+    r = _ref.ecref_matrix_rs_vandermonde_jerasure_w16(
+        a.ctypes.data_as(ctypes.c_void_p), k, m)
+    if r != 0:
+        raise ValueError(f"matrix_w16 failed: {r}")
+    return a
+
+
+def encode_w16(k, m, data, chunk_bytes=None):
+    """jerasure reed_sol_van w=16 (GF(2^16), u16 LE symbols)."""
+    lens = {d.nbytes for d in data if d is not None}
+    assert len(lens) == 1 or (not lens and chunk_bytes)
+    length = lens.pop() if lens else chunk_bytes
+    rows = np.ascontiguousarray(matrix_w16(k, m)[k:])
+    parity = [np.zeros(length, dtype=np.uint8) for _ in range(m)]
+    _ref.ecref_encode16(k, m, rows.ctypes.data_as(ctypes.c_void_p),
+                        _ptr_array(data), _ptr_array(parity),
+                        ctypes.c_size_t(length))
+    return parity
+
+
+def decode_w16(k, m, chunks, present):
+    pres = np.asarray(present, dtype=np.uint8)
+    r = _ref.ecref_decode16(_ptr_array(chunks),
+                            pres.ctypes.data_as(ctypes.c_void_p), k, m,
+                            ctypes.c_size_t(chunks[0].nbytes))
+    if r != 0:
+        raise ValueError(f"decode_w16 failed: {r}")
     return chunks
 
 

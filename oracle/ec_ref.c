@@ -399,6 +399,265 @@ unsigned ecref_chunk_size_jerasure(int k, int w, unsigned stripe_width)
   return padded / k;
 }
 
+/* ---------------- GF(2^16) (w=16 jerasure RS-van) ---------------- */
+
+static uint16_t *gf16_log = NULL;  /* 65536 u16 */
+static uint16_t *gf16_exp = NULL;
+static int gf16_ready = 0;
+
+void ecref_gf16_init(void)
+{
+  if (gf16_ready)
+    return;
+  gf16_log = (uint16_t *)malloc(65536 * 2);
+  gf16_exp = (uint16_t *)malloc(65536 * 2);
+  unsigned v = 1;
+  for (int i = 0; i < 65535; i++) {
+    gf16_exp[i] = (uint16_t)v;
+    gf16_log[v] = (uint16_t)i;
+    v <<= 1;
+    if (v & 0x10000)
+      v ^= 0x1100B; /* gf-complete w=16 default polynomial */
+  }
+  gf16_exp[65535] = gf16_exp[0];
+  gf16_log[0] = 0;
+  gf16_ready = 1;
+}
+
+uint16_t ecref_gf16_mul(uint16_t a, uint16_t b)
+{
+  if (!a || !b)
+    return 0;
+  int s = gf16_log[a] + gf16_log[b];
+  if (s >= 65535)
+    s -= 65535;
+  return gf16_exp[s];
+}
+
+uint16_t ecref_gf16_inv(uint16_t a)
+{
+  if (!a)
+    return 0;
+  return gf16_exp[65535 - gf16_log[a]];
+}
+
+static uint16_t gf16_div(uint16_t a, uint16_t b)
+{
+  if (!a || !b)
+    return 0;
+  int s = gf16_log[a] - gf16_log[b];
+  if (s < 0)
+    s += 65535;
+  return gf16_exp[s];
+}
+
+/* same jerasure big-Vandermonde algorithm as the w=8 restatement above,
+ * in GF(2^16) */
+int ecref_matrix_rs_vandermonde_jerasure_w16(uint16_t *a, int k, int m)
+{
+  ecref_gf16_init();
+  int rows = k + m, cols = k;
+  if (k < 1 || m < 0 || rows > 65535)
+    return -EINVAL;
+  for (int j = 0; j < cols; j++)
+    a[j] = (j == 0);
+  if (rows > 1)
+    for (int j = 0; j < cols; j++)
+      a[(size_t)(rows - 1) * cols + j] = (j == cols - 1);
+  for (int i = 1; i < rows - 1; i++) {
+    uint16_t v = 1;
+    for (int j = 0; j < cols; j++) {
+      a[(size_t)i * cols + j] = v;
+      v = ecref_gf16_mul(v, (uint16_t)i);
+    }
+  }
+  for (int i = 1; i < cols; i++) {
+    int j = i;
+    while (j < rows && a[(size_t)j * cols + i] == 0)
+      j++;
+    if (j >= rows)
+      return -EDOM;
+    if (j != i)
+      for (int c = 0; c < cols; c++) {
+        uint16_t t = a[(size_t)j * cols + c];
+        a[(size_t)j * cols + c] = a[(size_t)i * cols + c];
+        a[(size_t)i * cols + c] = t;
+      }
+    uint16_t piv = a[(size_t)i * cols + i];
+    if (piv != 1) {
+      uint16_t inv = gf16_div(1, piv);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + i] = ecref_gf16_mul(inv, a[(size_t)r * cols + i]);
+    }
+    for (int c = 0; c < cols; c++) {
+      uint16_t t = a[(size_t)i * cols + c];
+      if (c != i && t != 0)
+        for (int r = 0; r < rows; r++)
+          a[(size_t)r * cols + c] ^=
+              ecref_gf16_mul(t, a[(size_t)r * cols + i]);
+    }
+  }
+  for (int j = 0; j < cols; j++) {
+    uint16_t t = a[(size_t)cols * cols + j];
+    if (t != 0 && t != 1) {
+      uint16_t inv = gf16_div(1, t);
+      for (int r = 0; r < rows; r++)
+        a[(size_t)r * cols + j] = ecref_gf16_mul(inv, a[(size_t)r * cols + j]);
+      for (int c = 0; c < cols; c++)
+        a[(size_t)j * cols + c] = ecref_gf16_mul(t, a[(size_t)j * cols + c]);
+    }
+  }
+  return 0;
+}
+
+void ecref_encode16(int k, int m, const uint16_t *coding_rows,
+                    const uint8_t *const *data, uint8_t *const *parity,
+                    size_t len)
+{
+  ecref_gf16_init();
+  size_t n = len / 2;
+  for (int j = 0; j < m; j++) {
+    uint16_t *out = (uint16_t *)parity[j];
+    memset(out, 0, len);
+    for (int i = 0; i < k; i++) {
+      if (data[i] == NULL)
+        continue;
+      uint16_t c = coding_rows[(size_t)j * k + i];
+      if (c == 0)
+        continue;
+      const uint16_t *d = (const uint16_t *)data[i];
+      if (c == 1) {
+        for (size_t b = 0; b < n; b++)
+          out[b] ^= d[b];
+      } else {
+        uint32_t lc = gf16_log[c];
+        for (size_t b = 0; b < n; b++) {
+          uint16_t v = d[b];
+          if (v) {
+            unsigned s = lc + gf16_log[v];
+            if (s >= 65535)
+              s -= 65535;
+            out[b] ^= gf16_exp[s];
+          }
+        }
+      }
+    }
+  }
+}
+
+static int gf16_invert_matrix(uint16_t *in, uint16_t *out, int k)
+{
+  memset(out, 0, (size_t)k * k * 2);
+  for (int i = 0; i < k; i++)
+    out[(size_t)i * k + i] = 1;
+  for (int i = 0; i < k; i++) {
+    if (!in[(size_t)i * k + i]) {
+      int j = i + 1;
+      while (j < k && !in[(size_t)j * k + i])
+        j++;
+      if (j >= k)
+        return -1;
+      for (int c = 0; c < k; c++) {
+        uint16_t t = in[(size_t)i * k + c];
+        in[(size_t)i * k + c] = in[(size_t)j * k + c];
+        in[(size_t)j * k + c] = t;
+        t = out[(size_t)i * k + c];
+        out[(size_t)i * k + c] = out[(size_t)j * k + c];
+        out[(size_t)j * k + c] = t;
+      }
+    }
+    uint16_t inv = ecref_gf16_inv(in[(size_t)i * k + i]);
+    for (int c = 0; c < k; c++) {
+      in[(size_t)i * k + c] = ecref_gf16_mul(inv, in[(size_t)i * k + c]);
+      out[(size_t)i * k + c] = ecref_gf16_mul(inv, out[(size_t)i * k + c]);
+    }
+    for (int r = 0; r < k; r++) {
+      if (r == i)
+        continue;
+      uint16_t f = in[(size_t)r * k + i];
+      if (!f)
+        continue;
+      for (int c = 0; c < k; c++) {
+        in[(size_t)r * k + c] ^= ecref_gf16_mul(f, in[(size_t)i * k + c]);
+        out[(size_t)r * k + c] ^= ecref_gf16_mul(f, out[(size_t)i * k + c]);
+      }
+    }
+  }
+  return 0;
+}
+
+int ecref_decode16(uint8_t *const *chunks, const uint8_t *present,
+                   int k, int m, size_t len)
+{
+  ecref_gf16_init();
+  uint16_t *gen = (uint16_t *)malloc((size_t)(k + m) * k * 2);
+  if (!gen)
+    return -ENOMEM;
+  if (ecref_matrix_rs_vandermonde_jerasure_w16(gen, k, m) != 0) {
+    free(gen);
+    return -1;
+  }
+  int n = k + m, nerrs = 0, erasures[255], decode_index[255];
+  for (int i = 0; i < n; i++)
+    if (!present[i])
+      erasures[nerrs++] = i;
+  if (nerrs == 0) {
+    free(gen);
+    return 0;
+  }
+  if (nerrs > m) {
+    free(gen);
+    return -1;
+  }
+  {
+    int r = 0;
+    for (int i = 0; i < k; i++, r++) {
+      while (r < n && !present[r])
+        r++;
+      if (r >= n) {
+        free(gen);
+        return -1;
+      }
+      decode_index[i] = r;
+    }
+  }
+  uint16_t *b = (uint16_t *)malloc((size_t)k * k * 2);
+  uint16_t *d = (uint16_t *)malloc((size_t)k * k * 2);
+  uint16_t *c = (uint16_t *)malloc((size_t)nerrs * k * 2);
+  if (!b || !d || !c) {
+    free(gen); free(b); free(d); free(c);
+    return -ENOMEM;
+  }
+  for (int i = 0; i < k; i++)
+    memcpy(&b[(size_t)i * k], &gen[(size_t)decode_index[i] * k], k * 2);
+  if (gf16_invert_matrix(b, d, k) != 0) {
+    free(gen); free(b); free(d); free(c);
+    return -1;
+  }
+  for (int p = 0; p < nerrs; p++) {
+    if (erasures[p] < k) {
+      memcpy(&c[(size_t)p * k], &d[(size_t)erasures[p] * k], k * 2);
+    } else {
+      for (int i = 0; i < k; i++) {
+        uint16_t s = 0;
+        for (int j = 0; j < k; j++)
+          s ^= ecref_gf16_mul(d[(size_t)j * k + i],
+                              gen[(size_t)erasures[p] * k + j]);
+        c[(size_t)p * k + i] = s;
+      }
+    }
+  }
+  const uint8_t *src[255];
+  uint8_t *dst[255];
+  for (int i = 0; i < k; i++)
+    src[i] = chunks[decode_index[i]];
+  for (int p = 0; p < nerrs; p++)
+    dst[p] = chunks[erasures[p]];
+  ecref_encode16(k, nerrs, c, src, dst, len);
+  free(gen); free(b); free(d); free(c);
+  return 0;
+}
+
 /* ---------------- jerasure bitmatrix (Cauchy-original) family ----------- */
 
 /* cauchy.c cauchy_original_coding_matrix: m[i][j] = 1/(i XOR (m+j)). */

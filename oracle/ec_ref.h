@@ -113,6 +113,24 @@ void ecref_region_mul_xor(uint8_t coeff, const uint8_t *delta, uint8_t *parity,
 unsigned ecref_chunk_size_isa(int k, unsigned stripe_width);       /* align 32 */
 unsigned ecref_chunk_size_jerasure(int k, int w, unsigned stripe_width);
 
+/* ---- GF(2^16), w=16 jerasure reed_sol_van ----
+ * gf-complete's default w=16 field (poly 0x1100B, x^16+x^12+x^3+x+1;
+ * submodule absent — published constant) with the same jerasure
+ * big-Vandermonde systematic construction, coefficients and symbols u16
+ * little-endian (galois_w16_region_multiply semantics,
+ * ErasureCodeJerasure.cc:316-319). */
+void ecref_gf16_init(void);
+uint16_t ecref_gf16_mul(uint16_t a, uint16_t b);
+uint16_t ecref_gf16_inv(uint16_t a);
+/* full (k+m) x k generator, identity top, u16 entries */
+int  ecref_matrix_rs_vandermonde_jerasure_w16(uint16_t *a, int k, int m);
+/* len bytes (multiple of 2); data[i] NULL => zeros */
+void ecref_encode16(int k, int m, const uint16_t *coding_rows,
+                    const uint8_t *const *data, uint8_t *const *parity,
+                    size_t len);
+int  ecref_decode16(uint8_t *const *chunks, const uint8_t *present,
+                    int k, int m, size_t len);
+
 /* ---- jerasure bitmatrix (Cauchy-original) family, w=8 ----
  * Restates jerasure cauchy.c cauchy_original_coding_matrix (m[i][j] =
  * 1/(i XOR (m+j))), jerasure.c jerasure_matrix_to_bitmatrix (w x w
