@@ -70,6 +70,8 @@ def parity_selfcheck(ctx, dptr, args, seed, sample_stripes=2):
             raise AssertionError(f"stripe {s}: data region != expected fill")
         if args.technique == "cauchy_orig":
             want = oracle.bitmatrix_encode(k, m, data, 2048)
+        elif args.technique == "jerasure_reed_sol_van_w16":
+            want = oracle.encode_w16(k, m, data)
         else:
             want = oracle.encode(args.technique, k, m, data)
         for j in range(m):
@@ -492,7 +494,9 @@ def main():
             },
             "cpu_baseline": (cpu_baseline(args)
                              if (world == 1 and not args.no_cpu_baseline
-                                 and args.technique != "cauchy_orig")
+                                 and args.technique in
+                                 ("reed_sol_van", "cauchy",
+                                  "jerasure_reed_sol_van"))
                              else None),
         }
         del line["config"]["encode_gibs"]
