@@ -86,6 +86,13 @@ def _lib():
                                       ctypes.c_int, ctypes.c_uint64,
                                       ctypes.c_int]
     lib.ecx_set_matrix.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    lib.ecx_matmul_batch.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.c_long, ctypes.c_size_t,
+                                     ctypes.POINTER(ctypes.c_int),
+                                     ctypes.c_int,
+                                     ctypes.POINTER(ctypes.c_int),
+                                     ctypes.c_int, ctypes.c_void_p,
+                                     ctypes.c_int]
     lib.ecx_shec_matrix.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int,
                                     ctypes.c_int, ctypes.c_void_p]
     lib.ecx_sync.argtypes = [ctypes.c_void_p, ctypes.c_int]
@@ -240,6 +247,19 @@ class EcContext:
         arr, sz, n = self._slice_args(chunk_ptrs, sizes)
         _ck(lib().ecx_decode_slices(self._h, arr, sz, n, present_mask,
                                     slot), "ecx_decode_slices")
+
+    def matmul_batch(self, dptr, n_stripes, chunk_bytes, src_ids, out_ids,
+                     rows, slot=0):
+        """Generic device-batch GF matmul over chunk ids (LRC layer
+        composition primitive)."""
+        rows = np.ascontiguousarray(rows, dtype=np.uint8)
+        assert rows.shape == (len(out_ids), len(src_ids))
+        sa = (ctypes.c_int * len(src_ids))(*src_ids)
+        oa = (ctypes.c_int * len(out_ids))(*out_ids)
+        _ck(lib().ecx_matmul_batch(self._h, dptr, n_stripes, chunk_bytes,
+                                   sa, len(src_ids), oa, len(out_ids),
+                                   rows.ctypes.data_as(ctypes.c_void_p),
+                                   slot), "ecx_matmul_batch")
 
     def set_matrix(self, coding_rows):
         """Replace the coding rows (custom-matrix codecs, e.g. SHEC)."""

@@ -1574,6 +1574,23 @@ int ecx_shec_matrix(int k, int m, int c, int single, uint8_t* out) {
   return ECX_OK;
 }
 
+int ecx_matmul_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
+                     size_t chunk_bytes, const int* src_ids, int n_src,
+                     const int* out_ids, int n_out, const uint8_t* rows,
+                     int slot) {
+  if (!ctx || !dptr || !src_ids || !out_ids || !rows || ctx->is_bitmatrix() ||
+      ctx->is_w16())
+    return ECX_ERR_INVAL;
+  // chunk ids must lie within the batch stripe (k+m chunks)
+  for (int i = 0; i < n_src; i++)
+    if (src_ids[i] < 0 || src_ids[i] >= ctx->k + ctx->m) return ECX_ERR_INVAL;
+  for (int j = 0; j < n_out; j++)
+    if (out_ids[j] < 0 || out_ids[j] >= ctx->k + ctx->m) return ECX_ERR_INVAL;
+  return run_matmul(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr, src_ids,
+                    n_src, out_ids, n_out, rows, nullptr, n_stripes,
+                    chunk_bytes, false);
+}
+
 int ecx_sync(ecx_ctx* ctx, int slot) {
   if (!ctx || slot < 0 || slot >= (int)ctx->slots.size()) return ECX_ERR_INVAL;
   HIP_TRY(hipSetDevice(ctx->device));

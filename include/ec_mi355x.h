@@ -213,6 +213,15 @@ ECX_API int ecx_matmul_chunks_host(ecx_ctx *ctx,
  * single != 0 selects the SINGLE technique. */
 ECX_API int ecx_shec_matrix(int k, int m, int c, int single, uint8_t *out);
 
+/* Generic device-batch GF(2^8) matmul over the standard batch layout:
+ * out chunk ids = XOR_i rows[j*n_src+i] * src chunk ids, per stripe. The
+ * composition primitive for layered codes (LRC layers, custom research
+ * codes) on device-resident batches; w=8 matrix techniques only. */
+ECX_API int ecx_matmul_batch(ecx_ctx *ctx, void *dptr, long n_stripes,
+                             size_t chunk_bytes, const int *src_ids,
+                             int n_src, const int *out_ids, int n_out,
+                             const uint8_t *rows, int slot);
+
 /* Synchronise a stream slot. */
 ECX_API int ecx_sync(ecx_ctx *ctx, int slot);
 
