@@ -113,3 +113,106 @@ def test_golden_vectors():
                     for _ in range(k)]
             par = oracle.encode(tech, k, m, data)
             assert (np.stack(par) == gold[f"par_{key}"]).all()
+
+
+def _gf_inv_matrix(M, mul, inv):
+    """Gauss-Jordan inverse over a GF given mul/inv callables (test-local,
+    independent of the oracle's implementation)."""
+    n = M.shape[0]
+    A = M.copy()
+    I = np.eye(n, dtype=M.dtype)
+    for i in range(n):
+        if A[i, i] == 0:
+            j = next(r for r in range(i + 1, n) if A[r, i] != 0)
+            A[[i, j]] = A[[j, i]]
+            I[[i, j]] = I[[j, i]]
+        pinv = inv(int(A[i, i]))
+        for c in range(n):
+            A[i, c] = mul(pinv, int(A[i, c]))
+            I[i, c] = mul(pinv, int(I[i, c]))
+        for r in range(n):
+            if r == i or A[r, i] == 0:
+                continue
+            f = int(A[r, i])
+            for c in range(n):
+                A[r, c] ^= mul(f, int(A[i, c]))
+                I[r, c] ^= mul(f, int(I[i, c]))
+    return I
+
+
+def test_jerasure_vandermonde_uniqueness_crosscheck():
+    """The jerasure RS-van coding matrix is UNIQUELY determined by two
+    published facts, independent of elimination order: (1) the code of the
+    extended Vandermonde matrix (reed_sol_extended_vandermonde_matrix: row
+    0 = e0, rows 1..k+m-2 = powers of i, last row = e_{k-1}); (2) the
+    normalisation 'systematic top identity + first coding row all ones'.
+    Derivation: G_sys = G_ext * inv(G_ext_top); coding = P * diag(1/P[0]).
+    This cross-check computes that closed form independently and compares
+    with the oracle's elimination-based restatement — pinning the
+    construction beyond 'same algorithm transcribed'."""
+    for (k, m) in ((2, 1), (4, 3), (8, 3), (10, 4)):
+        # independent extended Vandermonde
+        n = k + m
+        G = np.zeros((n, k), dtype=np.uint8)
+        G[0, 0] = 1
+        G[n - 1, k - 1] = 1
+        for i in range(1, n - 1):
+            v = 1
+            for j in range(k):
+                G[i, j] = v
+                v = oracle.gf_mul(v, i)
+        top_inv = _gf_inv_matrix(G[:k], oracle.gf_mul, oracle.gf_inv)
+        # G_sys = G @ top_inv over GF(2^8)
+        P = np.zeros((m, k), dtype=np.uint8)
+        for r in range(m):
+            for c in range(k):
+                s = 0
+                for t in range(k):
+                    s ^= oracle.gf_mul(int(G[k + r, t]), int(top_inv[t, c]))
+                P[r, c] = s
+        assert (P[0] != 0).all()
+        for c in range(k):
+            sc = oracle.gf_inv(int(P[0, c]))
+            for r in range(m):
+                P[r, c] = oracle.gf_mul(int(P[r, c]), sc)
+        got = oracle.matrix("jerasure_reed_sol_van", k, m)[k:]
+        assert np.array_equal(got, P), (k, m)
+
+
+def test_jerasure_vandermonde_w16_uniqueness_crosscheck():
+    """Same closed-form cross-check in GF(2^16)."""
+    def inv16(a):
+        # brute via log tables through oracle.gf16_mul search is slow;
+        # use Fermat: a^(2^16-2)
+        r, e, b = 1, 65534, a
+        while e:
+            if e & 1:
+                r = oracle.gf16_mul(r, b)
+            b = oracle.gf16_mul(b, b)
+            e >>= 1
+        return r
+
+    k, m = 5, 3
+    n = k + m
+    G = np.zeros((n, k), dtype=np.uint16)
+    G[0, 0] = 1
+    G[n - 1, k - 1] = 1
+    for i in range(1, n - 1):
+        v = 1
+        for j in range(k):
+            G[i, j] = v
+            v = oracle.gf16_mul(v, i)
+    top_inv = _gf_inv_matrix(G[:k], oracle.gf16_mul, inv16)
+    P = np.zeros((m, k), dtype=np.uint16)
+    for r in range(m):
+        for c in range(k):
+            s = 0
+            for t in range(k):
+                s ^= oracle.gf16_mul(int(G[k + r, t]), int(top_inv[t, c]))
+            P[r, c] = s
+    for c in range(k):
+        sc = inv16(int(P[0, c]))
+        for r in range(m):
+            P[r, c] = oracle.gf16_mul(int(P[r, c]), sc)
+    got = oracle.matrix_w16(k, m)[k:]
+    assert np.array_equal(got, P)
