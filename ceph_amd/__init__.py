@@ -99,6 +99,7 @@ def _lib():
     lib.ecx_last_kernel_ms.argtypes = [ctypes.c_void_p, ctypes.c_int,
                                        ctypes.POINTER(ctypes.c_double)]
     lib.ecx_get_matrix.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    lib.ecx_get_matrix16.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
     lib.ecx_encode_chunks_host.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                            ctypes.c_void_p, ctypes.c_size_t]
     lib.ecx_decode_chunks_host.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
@@ -178,6 +179,12 @@ class EcContext:
             pass
 
     def matrix(self):
+        if TECHNIQUES.get(self.technique) == T_RS_VAN_JERASURE_W16:
+            a = np.zeros((self.k + self.m, self.k), dtype=np.uint16)
+            _ck(lib().ecx_get_matrix16(
+                self._h, a.ctypes.data_as(ctypes.c_void_p)),
+                "ecx_get_matrix16")
+            return a
         a = np.zeros((self.k + self.m, self.k), dtype=np.uint8)
         _ck(lib().ecx_get_matrix(self._h, a.ctypes.data_as(ctypes.c_void_p)),
             "ecx_get_matrix")
