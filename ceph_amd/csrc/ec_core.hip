@@ -446,96 +446,59 @@ template <bool NT>
 __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
-    int windows_per_sw, int n_tiles) {
-  // PMC showed the single stage->barrier->compute shape parked 79% of
-  // wave cycles at waits: one window is ~14 KB and ~1.5 compute items per
-  // thread, so every wave just waited on HBM. This version grid-strides
-  // over window tiles with a double-buffered LDS stage: the next window's
-  // global loads are issued into registers while the current window is
-  // computed, hiding the load latency behind compute.
+    int windows_per_sw) {
   const EcBitParams* bp = (const EcBitParams*)blob;
   const uint16_t* g_ops = (const uint16_t*)(blob + sizeof(EcBitParams));
   extern __shared__ uint8_t smem[];
   const int n_src = bp->n_src, w = bp->w, pkt = bp->pkt, q = bp->q;
   const int vq = q >> 4, vq_shift = bp->vq_shift;
   const int n_rows = bp->n_out * w;
-  const int stage_bytes = n_src * w * q;  // <= 16 KB (host-enforced)
-  uint8_t* s_buf[2] = {smem, smem + stage_bytes};
-  uint16_t* s_ops = (uint16_t*)(smem + 2 * stage_bytes);
+  uint8_t* s_data = smem;                       // n_src*w*q bytes
+  uint16_t* s_ops = (uint16_t*)(smem + (size_t)n_src * w * q);
   const int n_ops = bp->row_off[n_rows];
   for (int t = threadIdx.x; t < n_ops; t += blockDim.x) s_ops[t] = g_ops[t];
 
+  const long tile = blockIdx.x;
+  const int win = (int)(tile % windows_per_sw);
+  const long sw = tile / windows_per_sw;
   const uint8_t* sbase = buf + (long)blockIdx.y * cps * chunk_bytes;
   uint8_t* obase = obuf + (long)blockIdx.y * cps * chunk_bytes;
-  const int stage_vecs = stage_bytes >> 4;
+  const long sw_off = sw * (long)w * pkt + (long)win * q;
 
-  // per-thread staging registers: ceil(stage_vecs/256) <= 4
-  v4u sr[4];
-  auto issue_loads = [&](int tile) {
-    const int win = tile % windows_per_sw;
-    const long sw = tile / windows_per_sw;
-    const long sw_off = sw * (long)w * pkt + (long)win * q;
-#pragma unroll
-    for (int i = 0; i < 4; i++) {
-      const int t = threadIdx.x + i * 256;
-      if (t >= stage_vecs) break;
-      const int jc = t >> vq_shift;
-      const int v = t - (jc << vq_shift);
-      const int j = jc / w, c = jc - j * w;
-      const v4u* src = reinterpret_cast<const v4u*>(
-          sbase + (long)bp->src_ids[j] * chunk_bytes + sw_off +
-          (long)c * pkt + (long)v * 16);
-      sr[i] = NT ? __builtin_nontemporal_load(src) : *src;
-    }
-  };
-  auto store_stage = [&](int half) {
-#pragma unroll
-    for (int i = 0; i < 4; i++) {
-      const int t = threadIdx.x + i * 256;
-      if (t >= stage_vecs) break;
-      *reinterpret_cast<v4u*>(s_buf[half] + (size_t)t * 16) = sr[i];
-    }
-  };
-
-  int tile = blockIdx.x;
-  if (tile < n_tiles) {
-    issue_loads(tile);
-    store_stage(0);
+  for (int t = threadIdx.x; t < n_src * w * vq; t += blockDim.x) {
+    const int jc = t >> vq_shift;
+    const int v = t - (jc << vq_shift);
+    const int j = jc / w, c = jc - j * w;
+    const v4u* src = reinterpret_cast<const v4u*>(
+        sbase + (long)bp->src_ids[j] * chunk_bytes + sw_off + (long)c * pkt +
+        (long)v * 16);
+    const v4u d = NT ? __builtin_nontemporal_load(src) : *src;
+    *reinterpret_cast<v4u*>(s_data + (size_t)jc * q + (size_t)v * 16) = d;
   }
-  int half = 0;
-  for (; tile < n_tiles; tile += gridDim.x) {
-    const int next = tile + gridDim.x;
-    if (next < n_tiles) issue_loads(next);  // overlap with compute below
-    __syncthreads();  // current half's stores visible; prior reads done
+  __syncthreads();
 
-    const int win = tile % windows_per_sw;
-    const long sw = tile / windows_per_sw;
-    const uint8_t* sd = s_buf[half];
-    for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
-      const int r = t >> vq_shift;
-      const int v = t - (r << vq_shift);
-      v4u acc = {0, 0, 0, 0};
-      const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
-      for (int o = b0; o < b1; o++) {
-        const int jc = s_ops[o];
-        const v4u d = *reinterpret_cast<const v4u*>(
-            sd + (size_t)jc * q + (size_t)v * 16);
-        acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
-      }
-      v4u* dst = reinterpret_cast<v4u*>(
-          obase + (long)bp->out_ids[r / w] * chunk_bytes +
-          sw * (long)w * pkt + (long)(r % w) * pkt + (long)win * q +
-          (long)v * 16);
-      if (NT)
-        __builtin_nontemporal_store(acc, dst);
-      else
-        *dst = acc;
+  // Item-parallel compute (one 16B vec of one output row per item): A/B
+  // showed this beats a row-per-wave readlane variant — with q=512 the
+  // row-per-wave form idles half of each wave (vq=32) and larger q
+  // collapses residency; LDS op reads broadcast cheaply.
+  for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
+    const int r = t >> vq_shift;
+    const int v = t - (r << vq_shift);
+    v4u acc = {0, 0, 0, 0};
+    const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
+    for (int o = b0; o < b1; o++) {
+      const int jc = s_ops[o];
+      const v4u d = *reinterpret_cast<const v4u*>(
+          s_data + (size_t)jc * q + (size_t)v * 16);
+      acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
     }
-    if (next < n_tiles) {
-      __syncthreads();  // everyone done reading the other half
-      store_stage(half ^ 1);
-      half ^= 1;
-    }
+    v4u* dst = reinterpret_cast<v4u*>(
+        obase + (long)bp->out_ids[r / w] * chunk_bytes + sw * (long)w * pkt +
+        (long)(r % w) * pkt + (long)win * q + (long)v * 16);
+    if (NT)
+      __builtin_nontemporal_store(acc, dst);
+    else
+      *dst = acc;
   }
 }
 
@@ -1163,13 +1126,15 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
     return v ? atoi(v) : 1;
   }();
 
-  // LDS stage window (ECX_BITQ KB, hard-capped at 16 KB = the kernel's
-  // 4-registers-per-thread staging limit). Double-buffered in the kernel.
+  // LDS window: largest power-of-two divisor of pkt within the LDS budget
+  // (ECX_BITQ KB). Default 16 from the MI355X sweep: small windows keep
+  // 8+ blocks/CU resident and beat large windows by ~19%
+  // (profiles/rocprof_r01_summary.md).
   static const size_t lds_budget = [] {
     const char* v = getenv("ECX_BITQ");
     long kb = v ? atol(v) : 16;
-    if (kb < 4) kb = 4;
-    if (kb > 16) kb = 16;
+    if (kb < 8) kb = 8;
+    if (kb > 120) kb = 120;
     return (size_t)kb * 1024;
   }();
   int q = 16;
@@ -1214,24 +1179,17 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
 
   const long sw_per_chunk = (long)(chunk_bytes / ((size_t)w * pkt));
   const int windows_per_sw = pkt / q;
-  const long n_tiles = sw_per_chunk * windows_per_sw;
-  // grid-stride over tiles: enough concurrent blocks for residency, each
-  // looping >= ~2 tiles so the double buffer actually pipelines
-  long want_blocks = std::max<long>(1, 4096 / n_stripes);
-  long gx = std::min<long>(n_tiles, want_blocks);
-  if (gx < 1) gx = 1;
-  dim3 grid((unsigned)gx, (unsigned)n_stripes);
-  size_t lds =
-      2 * (size_t)n_src * w * q + ((ops.size() * 2 + 15) & ~15ull);
+  dim3 grid((unsigned)(sw_per_chunk * windows_per_sw), (unsigned)n_stripes);
+  size_t lds = (size_t)n_src * w * q + ((ops.size() * 2 + 15) & ~15ull);
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
   if (env_nt)
     hipLaunchKernelGGL((ec_bitmatrix_kernel<true>), grid, dim3(256), lds,
                        s.stream, d_buf, d_obuf, s.d_jobs, (long)chunk_bytes,
-                       ctx->k + ctx->m, windows_per_sw, (int)n_tiles);
+                       ctx->k + ctx->m, windows_per_sw);
   else
     hipLaunchKernelGGL((ec_bitmatrix_kernel<false>), grid, dim3(256), lds,
                        s.stream, d_buf, d_obuf, s.d_jobs, (long)chunk_bytes,
-                       ctx->k + ctx->m, windows_per_sw, (int)n_tiles);
+                       ctx->k + ctx->m, windows_per_sw);
   HIP_TRY(hipGetLastError());
   if (time_it) {
     HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
