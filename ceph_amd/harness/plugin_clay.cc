@@ -107,10 +107,17 @@ class ErasureCodeClay final : public ErasureCode {
     } else if (scalar == "jerasure") {
       sub_plugin = "mi355x";
       if (technique == "reed_sol_van") sub_tech = "jerasure_reed_sol_van";
+    } else if (scalar == "shec") {
+      // reference clay accepts shec sub-codecs (technique single/multiple,
+      // c forced to 2 — ErasureCodeClay.cc:344-347)
+      sub_plugin = "shec";
+      if (sub_tech != "single" && sub_tech != "multiple")
+        sub_tech = "single";
     } else if (scalar != "mi355x" && scalar != "oracle") {
       if (ss)
         *ss << "scalar_mds " << scalar
-            << " not supported here; use mi355x, oracle, isa or jerasure\n";
+            << " not supported here; use mi355x, oracle, shec, isa or "
+               "jerasure\n";
       return -EINVAL;
     }
 
@@ -120,6 +127,7 @@ class ErasureCodeClay final : public ErasureCode {
                             {"k", std::to_string(k_ + nu_)},
                             {"m", std::to_string(m_)},
                             {"w", "8"}};
+    if (sub_plugin == "shec") mdsp["c"] = "2";
     int r = registry.factory(sub_plugin, directory_, mdsp, &mds_, ss);
     if (r) return r;
     ErasureCodeProfile pftp{{"plugin", sub_plugin},
@@ -127,6 +135,7 @@ class ErasureCodeClay final : public ErasureCode {
                             {"k", "2"},
                             {"m", "2"},
                             {"w", "8"}};
+    if (sub_plugin == "shec") pftp["c"] = "2";
     r = registry.factory(sub_plugin, directory_, pftp, &pft_, ss);
     if (r) return r;
     return ErasureCode::init(profile, ss);

@@ -94,3 +94,13 @@ def test_chunk_size_alignment_rule():
                   "-s", "10000", "-i", "1", "-w", "decode", "-e", "1",
                   "-E", "exhaustive")
     assert r.returncode == 0, r.stderr + r.stdout
+
+
+@pytest.mark.gpu
+def test_clay_over_shec_subcodec():
+    """Reference clay accepts shec sub-codecs (ErasureCodeClay.cc:249-253,
+    344-347, c forced to 2); exhaustive 1-erasure decode byte-verified."""
+    r = run_bench("-p", "clay", "-P", "scalar_mds=shec", "-P", "k=4",
+                  "-P", "m=2", "-P", "d=5", "-s", str(4 * 32 * 1024),
+                  "-i", "1", "-w", "decode", "-e", "1", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
