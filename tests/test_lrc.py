@@ -79,3 +79,17 @@ def test_explicit_layers_profile():
                   "-s", "65536", "-i", "2", "-w", "decode", "-e", "1",
                   "-E", "exhaustive")
     assert r.returncode == 0, r.stderr
+
+
+def test_layers_with_object_profiles():
+    """The layers array's second element may be a JSON object
+    (ErasureCodeLrc.cc:176-201); exercise the object form with per-layer
+    plugin overrides."""
+    r = run_bench("-p", "lrc",
+                  "-P", "mapping=DD_DD_",
+                  "-P", 'layers=[ [ "DDcDDc", {"plugin": "oracle"} ], '
+                        '[ "DDc___", {"plugin": "oracle"} ], '
+                        '[ "___DDc", {"plugin": "oracle"} ] ]',
+                  "-s", "65536", "-i", "2", "-w", "decode", "-e", "1",
+                  "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
