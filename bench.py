@@ -150,11 +150,12 @@ def bench_mixed(args):
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    device = local_rank % ceph_amd.device_count()
     dist = None
     if world > 1:
         import torch
         import torch.distributed as tdist
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(device)
         tdist.init_process_group("nccl")
         dist = tdist
 
@@ -166,7 +167,7 @@ def bench_mixed(args):
     ctxs = []
     for (k, m, C) in shapes:
         S = max(8, int(2 * GIB // (k * C)))
-        ctx = ceph_amd.EcContext(k, m, "reed_sol_van", device=local_rank,
+        ctx = ceph_amd.EcContext(k, m, "reed_sol_van", device=device,
                                  n_streams=args.streams)
         nbytes = S * (k + m) * C
         d = ctx.dbuf_alloc(nbytes)
@@ -487,11 +488,14 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    # modulo so an N-rank run also works on fewer devices (single-box NCCL
+    # smoke tests); on the 8-GPU node this is the identity map
+    device = local_rank % ceph_amd.device_count()
     dist = None
     if world > 1:
         import torch
         import torch.distributed as tdist
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(device)
         tdist.init_process_group("nccl")
         dist = tdist
 
@@ -500,7 +504,7 @@ def main():
     buf_bytes = S * n * C
     seed = args.seed + rank
 
-    ctx = ceph_amd.EcContext(k, m, args.technique, device=local_rank,
+    ctx = ceph_amd.EcContext(k, m, args.technique, device=device,
                              n_streams=args.streams)
     dptr = ctx.dbuf_alloc(buf_bytes)
     ctx.fill_random(dptr, buf_bytes, seed)
