@@ -141,23 +141,15 @@ def cpu_baseline(args, budget_s=10.0):
     }
 
 
-def bench_mixed(args):
+def bench_mixed(args, dist=None, device=0):
     """BASELINE configs[4]: mixed k/m and 64 KiB..4 MiB chunks, streamed
     encode+decode with a per-(sub-batch) latency histogram. Shards across
     ranks like the main bench (weak scaling: every rank runs the full shape
-    set on its own GPU)."""
+    set on its own GPU). dist/device come from main() which handles the
+    torch-before-ceph_amd runtime ordering."""
     import ceph_amd
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
-    device = local_rank % ceph_amd.device_count()
-    dist = None
-    if world > 1:
-        import torch
-        import torch.distributed as tdist
-        torch.cuda.set_device(device)
-        tdist.init_process_group("nccl")
-        dist = tdist
 
     # (k, m, chunk_bytes): BASELINE's k in {4,6,8,12}, m in {2,3,4},
     # chunks 64 KiB..4 MiB; sub-batch sized ~2 GiB of data each
@@ -469,35 +461,42 @@ def main():
         args.k, args.m, args.technique, args.erasures = 10, 4, "cauchy", 4
         args.stripes = min(args.stripes, 3072)  # 14 chunks/stripe, ~42 GiB
 
+    # distributed setup (torchrun provides RANK/WORLD_SIZE/LOCAL_RANK).
+    # ORDER MATTERS: when torch is needed, its (bundled) HIP runtime must
+    # initialise BEFORE ceph_amd loads /opt/rocm's — loading ours first
+    # leaves torch.cuda seeing zero devices (two HIP runtimes in one
+    # process). ceph_amd then resolves libamdhip64 by soname to the
+    # already-loaded copy.
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    device = local_rank
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        # modulo so an N-rank run also works on fewer devices (single-box
+        # smoke tests); identity on the 8-GPU node
+        device = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(device)
+        tdist.init_process_group("nccl")
+        dist = tdist
+
     import ceph_amd
 
     if ceph_amd.device_count() < 1:
         print(json.dumps({"error": "no GPU visible; bench requires MI355X"}))
         sys.exit(1)
+    device = device % ceph_amd.device_count()
 
     if args.config == "mixed":
-        return bench_mixed(args)
+        return bench_mixed(args, dist, device)
     if args.config == "hostpath":
         return bench_hostpath(args)
     if args.config == "slices":
         return bench_slices(args)
     if args.config == "lrc":
         return bench_lrc(args)
-
-    # distributed setup (torchrun provides RANK/WORLD_SIZE/LOCAL_RANK)
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
-    # modulo so an N-rank run also works on fewer devices (single-box NCCL
-    # smoke tests); on the 8-GPU node this is the identity map
-    device = local_rank % ceph_amd.device_count()
-    dist = None
-    if world > 1:
-        import torch
-        import torch.distributed as tdist
-        torch.cuda.set_device(device)
-        tdist.init_process_group("nccl")
-        dist = tdist
 
     k, m, C, S = args.k, args.m, args.chunk_bytes, args.stripes
     n = k + m
