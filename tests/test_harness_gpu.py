@@ -87,3 +87,31 @@ def test_clay_flags_gpu():
                   "--flags")
     assert r.returncode == 0, r.stderr
     assert r.stdout.strip() == "partialread,requiresubchunks"
+
+
+def test_shec_decode_exhaustive_gpu():
+    """SHEC end-to-end on the GPU path (plugin_shec drives the mi355x
+    kernels via ecx_set_matrix): exhaustive single-erasure decode with
+    byte verification."""
+    r = run_bench("-p", "shec", "-P", "k=4", "-P", "m=3", "-P", "c=2",
+                  "-s", str(4 * 65536), "-i", "2", "-w", "decode",
+                  "-e", "1", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
+
+
+def test_clay_decode_exhaustive_gpu():
+    """Clay with the default scalar_mds=mi355x sub-codec: exhaustive
+    m-erasure decode through decode_layered, verified byte-exact."""
+    r = run_bench("-p", "clay", "-P", "k=4", "-P", "m=2", "-P", "d=5",
+                  "-s", str(4 * 65536), "-i", "2", "-w", "decode",
+                  "-e", "2", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
+
+
+def test_lrc_decode_exhaustive_gpu():
+    """LRC with the default mi355x sub-plugin for every layer: exhaustive
+    single-erasure decode (each recovered by its local layer on GPU)."""
+    r = run_bench("-p", "lrc", "-P", "k=4", "-P", "m=2", "-P", "l=3",
+                  "-s", str(4 * 65536), "-i", "2", "-w", "decode",
+                  "-e", "1", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
