@@ -553,6 +553,14 @@ struct Slot {
   // host-path staging stripe buffer, grown on demand
   uint8_t* d_stage = nullptr;
   size_t stage_bytes = 0;
+  // pipelined host-path: pinned+device double buffer and resident
+  // per-group launch params (see pipelined_matmul_host below)
+  uint8_t* h_pipe = nullptr;  // pinned, 2 x cps x tile
+  uint8_t* d_pipe = nullptr;
+  size_t pipe_bytes = 0;
+  EcLaunchParams* h_pparams = nullptr;  // pinned, 8 groups
+  EcLaunchParams* d_pparams = nullptr;
+  hipEvent_t ev_pipe[2] = {nullptr, nullptr};
   double last_ms = -1.0;
   bool timed = false;
   std::recursive_mutex mu;
@@ -669,6 +677,8 @@ int ecx_create2(int k, int m, int technique, int w, int packetsize,
         hipEventCreate(&s.ev_stop) != hipSuccess ||
         hipEventCreate(&s.ev_param) != hipSuccess ||
         hipEventCreate(&s.ev_jobs) != hipSuccess ||
+        hipEventCreate(&s.ev_pipe[0]) != hipSuccess ||
+        hipEventCreate(&s.ev_pipe[1]) != hipSuccess ||
         hipHostMalloc(&s.h_params, sizeof(EcLaunchParams)) != hipSuccess ||
         hipMalloc(&s.d_params, sizeof(EcLaunchParams)) != hipSuccess) {
       ecx_destroy(ctx);
@@ -693,11 +703,17 @@ void ecx_destroy(ecx_ctx* ctx) {
     if (s.ev_stop) (void)hipEventDestroy(s.ev_stop);
     if (s.ev_param) (void)hipEventDestroy(s.ev_param);
     if (s.ev_jobs) (void)hipEventDestroy(s.ev_jobs);
+    if (s.ev_pipe[0]) (void)hipEventDestroy(s.ev_pipe[0]);
+    if (s.ev_pipe[1]) (void)hipEventDestroy(s.ev_pipe[1]);
     if (s.h_jobs) (void)hipHostFree(s.h_jobs);
     if (s.d_jobs) (void)hipFree(s.d_jobs);
     if (s.h_params) (void)hipHostFree(s.h_params);
     if (s.d_params) (void)hipFree(s.d_params);
     if (s.d_stage) (void)hipFree(s.d_stage);
+    if (s.h_pipe) (void)hipHostFree(s.h_pipe);
+    if (s.d_pipe) (void)hipFree(s.d_pipe);
+    if (s.h_pparams) (void)hipHostFree(s.h_pparams);
+    if (s.d_pparams) (void)hipFree(s.d_pparams);
     if (s.stream) (void)hipStreamDestroy(s.stream);
   }
   delete ctx;
@@ -853,24 +869,15 @@ static void fill_params(EcLaunchParams* p, const ecx::GF8& f,
     }
 }
 
-// Launch the matmul kernel for one output group (n_out <= ECX_MAX_OUT but
-// template-dispatched in groups of <= 4 for register economy).
-static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
-                         uint8_t* d_obuf, const EcLaunchParams& params,
-                         long n_stripes, size_t chunk_bytes, bool accum,
-                         bool time_it) {
-  if (chunk_bytes % 16 || n_stripes <= 0 || n_stripes > 65535)
-    return ECX_ERR_INVAL;
-  HIP_TRY(hipSetDevice(ctx->device));
+// Grid/variant selection for the matmul kernel, shared by the slot-staged
+// launch_matmul and the pipelined host path (which keeps params resident).
+struct MatmulCfg {
+  dim3 grid;
+  int vpt;
+  bool nt;
+};
 
-  // wait for any in-flight param upload on this slot, then stage params
-  HIP_TRY(hipEventSynchronize(s.ev_param));
-  std::memcpy(s.h_params, &params, sizeof(EcLaunchParams));
-  HIP_TRY(hipMemcpyAsync(s.d_params, s.h_params, sizeof(EcLaunchParams),
-                         hipMemcpyHostToDevice, s.stream));
-  HIP_TRY(hipEventRecord(s.ev_param, s.stream));
-
-  const long vecs = (long)(chunk_bytes >> 4);
+static MatmulCfg matmul_cfg(long vecs, long n_stripes) {
   // Tunables (A/B-able via env on the GPU box): VPT = 16B vectors per
   // thread per iteration, TILE = target vectors per thread per launch.
   // Defaults from the round-1 sweeps (profiles/rocprof_r01_summary.md):
@@ -893,18 +900,30 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
     const char* v = getenv("ECX_NT");
     return v ? atoi(v) : 1;
   }();
-  const int vpt = (vecs >= 2 * 256) ? env_vpt : 1;
-  const long per_block = 256L * vpt;
+  MatmulCfg c;
+  c.vpt = (vecs >= 2 * 256) ? env_vpt : 1;
+  const long per_block = 256L * c.vpt;
   long tiles = (vecs + per_block - 1) / per_block;
-  long loops = std::max(1, env_tile / vpt);
+  long loops = std::max(1, env_tile / c.vpt);
   int gx = (int)std::min<long>((tiles + loops - 1) / loops, 1024);
   if (gx < 1) gx = 1;
   // if few stripes, widen x so total blocks cover 256 CUs * a few waves
   while ((long)gx * n_stripes < 2048 && gx < tiles) gx *= 2;
-  dim3 grid(gx, (unsigned)n_stripes);
+  c.grid = dim3(gx, (unsigned)n_stripes);
+  c.nt = env_nt != 0;
+  return c;
+}
 
-  if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
-  const int cps = ctx->k + ctx->m;
+// Template dispatch for one output group (n_out <= 4) with params already
+// resident on the device.
+static int matmul_dispatch(hipStream_t stream, const uint8_t* d_buf,
+                           uint8_t* d_obuf, const EcLaunchParams* d_params,
+                           int n_out, int n_src, bool accum,
+                           const MatmulCfg& c, size_t chunk_bytes, int cps) {
+  const long vecs = (long)(chunk_bytes >> 4);
+  const dim3 grid = c.grid;
+  const int vpt = c.vpt;
+  const bool env_nt = c.nt;
   // K8 full source-unroll measured 2.1x SLOWER (16.7 vs 8.0 ms encode:
   // the unrolled body bloats registers/issue and the dynamic loop already
   // gets enough MLP from 8 waves/SIMD) — keep available for experiments,
@@ -913,14 +932,14 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
     const char* v = getenv("ECX_K8");
     return v ? atoi(v) : 0;
   }();
-  const bool k8 = env_k8 && !accum && vpt == 1 && params.n_src == 8;
+  const bool k8 = env_k8 && !accum && vpt == 1 && n_src == 8;
 #define ECX_LAUNCH(NO, AC, VP, NTF)                                          \
   hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF>), grid,           \
-                     dim3(256), 0, s.stream, d_buf, d_obuf, s.d_params,      \
+                     dim3(256), 0, stream, d_buf, d_obuf, d_params,          \
                      (long)chunk_bytes, cps, vecs)
 #define ECX_LAUNCH_K8(NO, NTF)                                               \
   hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, false, 1, NTF, 8>), grid,      \
-                     dim3(256), 0, s.stream, d_buf, d_obuf, s.d_params,      \
+                     dim3(256), 0, stream, d_buf, d_obuf, d_params,          \
                      (long)chunk_bytes, cps, vecs)
 #define ECX_VARIANT(NO, AC)                          \
   do {                                               \
@@ -940,7 +959,7 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
     if (accum) ECX_VARIANT(NO, true); \
     else ECX_VARIANT(NO, false);      \
     break;
-  switch (params.n_out) {
+  switch (n_out) {
     ECX_DISPATCH(1)
     ECX_DISPATCH(2)
     ECX_DISPATCH(3)
@@ -953,6 +972,32 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
 #undef ECX_LAUNCH_K8
 #undef ECX_LAUNCH
   HIP_TRY(hipGetLastError());
+  return ECX_OK;
+}
+
+// Launch the matmul kernel for one output group (n_out <= ECX_MAX_OUT but
+// template-dispatched in groups of <= 4 for register economy).
+static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
+                         uint8_t* d_obuf, const EcLaunchParams& params,
+                         long n_stripes, size_t chunk_bytes, bool accum,
+                         bool time_it) {
+  if (chunk_bytes % 16 || n_stripes <= 0 || n_stripes > 65535)
+    return ECX_ERR_INVAL;
+  HIP_TRY(hipSetDevice(ctx->device));
+
+  // wait for any in-flight param upload on this slot, then stage params
+  HIP_TRY(hipEventSynchronize(s.ev_param));
+  std::memcpy(s.h_params, &params, sizeof(EcLaunchParams));
+  HIP_TRY(hipMemcpyAsync(s.d_params, s.h_params, sizeof(EcLaunchParams),
+                         hipMemcpyHostToDevice, s.stream));
+  HIP_TRY(hipEventRecord(s.ev_param, s.stream));
+
+  const MatmulCfg c = matmul_cfg((long)(chunk_bytes >> 4), n_stripes);
+  if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
+  int r = matmul_dispatch(s.stream, d_buf, d_obuf, s.d_params, params.n_out,
+                          params.n_src, accum, c, chunk_bytes,
+                          ctx->k + ctx->m);
+  if (r != ECX_OK) return r;
   if (time_it) {
     HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
     s.timed = true;
@@ -1506,6 +1551,11 @@ int ecx_decode_slices(ecx_ctx* ctx, void* const* d_chunks,
 
 // defined in the host-pointer section below
 static int ensure_stage(ecx_ctx* ctx, Slot& s, size_t bytes);
+static int env_hostpipe();
+static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
+                                 const uint8_t* const* srcs, int n_src,
+                                 uint8_t* const* outs, int n_out,
+                                 const uint8_t* rows, size_t chunk_bytes);
 
 int ecx_set_matrix(ecx_ctx* ctx, const uint8_t* coding_rows) {
   if (!ctx || !coding_rows || ctx->is_bitmatrix() || ctx->is_w16())
@@ -1528,6 +1578,9 @@ int ecx_matmul_chunks_host(ecx_ctx* ctx, const uint8_t* const* srcs,
     return ECX_ERR_INVAL;
   Slot& s = ctx->slots[ctx->rr++ % ctx->slots.size()];
   std::lock_guard<std::recursive_mutex> g(s.mu);
+  if (env_hostpipe())
+    return pipelined_matmul_host(ctx, s, srcs, n_src, outs, n_out, rows,
+                                 bytes);
   int r = ensure_stage(ctx, s, (size_t)(n_src + n_out) * bytes);
   if (r != ECX_OK) return r;
   HIP_TRY(hipSetDevice(ctx->device));
@@ -1626,6 +1679,204 @@ static int ensure_stage(ecx_ctx* ctx, Slot& s, size_t bytes) {
   return ECX_OK;
 }
 
+// ---- pipelined host-pointer staging ---------------------------------------
+// The PCIe probe (tools/pcie_probe.py, profiles/rocprof_r01_summary.md)
+// measured 53 GiB/s per direction with pageable ~ pinned and no duplex
+// gain — but the legacy per-chunk pageable hipMemcpyAsync path pays ~50 us
+// of fixed cost PER COPY, so k+m copies per call capped the drop-in path at
+// 8.5 GiB/s (k=8 m=3, 1 MiB chunks). This path instead gathers the chunks
+// into a pinned double buffer with a parallel CPU memcpy (29 GiB/s per
+// core measured on the box), issues ONE H2D and ONE D2H DMA per stripe
+// tile, keeps the per-group launch params resident across tiles, and
+// overlaps the CPU gather/scatter of tile t with the DMA+kernel of tile
+// t-1. Knobs: ECX_HOSTPIPE=0 reverts to the legacy path, ECX_HPIPE_TILE
+// sets per-chunk tile bytes (default 256 KiB), ECX_HPIPE_THREADS the CPU
+// copy threads.
+
+static int env_hostpipe() {
+  static const int v = [] {
+    const char* e = getenv("ECX_HOSTPIPE");
+    return e ? atoi(e) : 1;
+  }();
+  return v;
+}
+
+static size_t env_hpipe_tile() {
+  static const size_t v = [] {
+    const char* e = getenv("ECX_HPIPE_TILE");
+    long x = e ? atol(e) : (256 << 10);
+    if (x < (16 << 10)) x = 16 << 10;
+    return (size_t)x & ~(size_t)15;
+  }();
+  return v;
+}
+
+struct CopyOp {
+  uint8_t* dst;
+  const uint8_t* src;
+  size_t n;
+};
+
+static void par_copy(const std::vector<CopyOp>& ops) {
+  size_t total = 0;
+  for (const auto& o : ops) total += o.n;
+#ifdef _OPENMP
+  if (total >= (256 << 10)) {
+    static const int nthr = [] {
+      const char* e = getenv("ECX_HPIPE_THREADS");
+      int x = e ? atoi(e) : 8;
+      return x < 1 ? 1 : (x > 64 ? 64 : x);
+    }();
+    constexpr size_t PIECE = 256 << 10;
+    std::vector<CopyOp> pieces;
+    pieces.reserve(ops.size() * 4);
+    for (const auto& o : ops)
+      for (size_t off = 0; off < o.n; off += PIECE)
+        pieces.push_back({o.dst + off, o.src + off,
+                          std::min(PIECE, o.n - off)});
+#pragma omp parallel for num_threads(nthr) schedule(static)
+    for (long i = 0; i < (long)pieces.size(); i++)
+      std::memcpy(pieces[i].dst, pieces[i].src, pieces[i].n);
+    return;
+  }
+#endif
+  for (const auto& o : ops) std::memcpy(o.dst, o.src, o.n);
+}
+
+static int ensure_pipe(ecx_ctx* ctx, Slot& s, size_t bytes) {
+  if (s.pipe_bytes >= bytes) return ECX_OK;
+  HIP_TRY(hipSetDevice(ctx->device));
+  HIP_TRY(hipEventSynchronize(s.ev_pipe[0]));
+  HIP_TRY(hipEventSynchronize(s.ev_pipe[1]));
+  if (s.h_pipe) (void)hipHostFree(s.h_pipe);
+  if (s.d_pipe) (void)hipFree(s.d_pipe);
+  s.h_pipe = nullptr;
+  s.d_pipe = nullptr;
+  s.pipe_bytes = 0;
+  HIP_TRY(hipHostMalloc(&s.h_pipe, bytes));
+  hipError_t e = hipMalloc(&s.d_pipe, bytes);
+  if (e != hipSuccess) {
+    (void)hipHostFree(s.h_pipe);
+    s.h_pipe = nullptr;
+    return map_hip(e);
+  }
+  s.pipe_bytes = bytes;
+  return ECX_OK;
+}
+
+static int ensure_pparams(ecx_ctx* ctx, Slot& s) {
+  if (s.h_pparams) return ECX_OK;
+  HIP_TRY(hipSetDevice(ctx->device));
+  HIP_TRY(hipHostMalloc(&s.h_pparams, 8 * sizeof(EcLaunchParams)));
+  hipError_t e = hipMalloc(&s.d_pparams, 8 * sizeof(EcLaunchParams));
+  if (e != hipSuccess) {
+    (void)hipHostFree(s.h_pparams);
+    s.h_pparams = nullptr;
+    return map_hip(e);
+  }
+  return ECX_OK;
+}
+
+// Generic host-pointer matmul (w=8 table techniques): srcs[i] == NULL is
+// the zeros-chunk convention (skipped via cls), outs[j] == NULL skips the
+// scatter of that output. Caller holds the slot lock.
+static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
+                                 const uint8_t* const* srcs, int n_src,
+                                 uint8_t* const* outs, int n_out,
+                                 const uint8_t* rows, size_t chunk_bytes) {
+  if (n_src < 1 || n_src > ECX_MAX_K || n_out < 1 || n_out > ECX_MAX_K ||
+      !chunk_bytes || chunk_bytes % 16)
+    return ECX_ERR_INVAL;
+  const int cps = n_src + n_out;
+  const size_t TS = std::min(env_hpipe_tile(), chunk_bytes);
+  int r = ensure_pipe(ctx, s, 2 * (size_t)cps * TS);
+  if (r != ECX_OK) return r;
+  r = ensure_pparams(ctx, s);
+  if (r != ECX_OK) return r;
+  HIP_TRY(hipSetDevice(ctx->device));
+
+  bool src_null[ECX_MAX_K] = {};
+  for (int i = 0; i < n_src; i++) src_null[i] = (srcs[i] == nullptr);
+  int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
+  for (int i = 0; i < n_src; i++) src_ids[i] = i;
+  for (int j = 0; j < n_out; j++) out_ids[j] = n_src + j;
+
+  // coefficient tables are tile-invariant: build and upload them ONCE,
+  // then every tile's kernels reference the resident copies
+  const ecx::GF8& f = ecx::gf8();
+  const int groups = (n_out + 3) / 4;
+  if (groups > 8) return ECX_ERR_INVAL;
+  // a previous successful call drained both pipe events, but an errored
+  // one may not have: make the pinned param area provably safe to rewrite
+  HIP_TRY(hipEventSynchronize(s.ev_pipe[0]));
+  HIP_TRY(hipEventSynchronize(s.ev_pipe[1]));
+  for (int g = 0; g < groups; g++) {
+    const int j0 = 4 * g, nj = std::min(4, n_out - j0);
+    fill_params(&s.h_pparams[g], f, src_ids, n_src, out_ids + j0, nj,
+                rows + (size_t)j0 * n_src, src_null);
+  }
+  HIP_TRY(hipMemcpyAsync(s.d_pparams, s.h_pparams,
+                         (size_t)groups * sizeof(EcLaunchParams),
+                         hipMemcpyHostToDevice, s.stream));
+
+  const long T = (long)((chunk_bytes + TS - 1) / TS);
+  size_t tl_of[2] = {0, 0}, off_of[2] = {0, 0};
+  std::vector<CopyOp> ops;
+  ops.reserve(cps);
+  for (long t = 0; t < T; t++) {
+    const int b = (int)(t & 1);
+    const size_t off = (size_t)t * TS;
+    const size_t tl = std::min(TS, chunk_bytes - off);
+    uint8_t* hbuf = s.h_pipe + (size_t)b * cps * TS;
+    uint8_t* dbuf = s.d_pipe + (size_t)b * cps * TS;
+    // buffer b holds tile t-2 until its D2H lands; wait, scatter it out,
+    // then refill — the CPU work here overlaps tile t-1's DMA + kernels
+    HIP_TRY(hipEventSynchronize(s.ev_pipe[b]));
+    if (t >= 2) {
+      ops.clear();
+      for (int j = 0; j < n_out; j++)
+        if (outs[j])
+          ops.push_back({outs[j] + off_of[b],
+                         hbuf + ((size_t)n_src + j) * tl_of[b], tl_of[b]});
+      par_copy(ops);
+    }
+    // gather tile t's sources packed at tl spacing (gaps where srcs[i] is
+    // NULL are never read: cls==0 skips those sources in the kernel)
+    ops.clear();
+    for (int i = 0; i < n_src; i++)
+      if (srcs[i]) ops.push_back({hbuf + (size_t)i * tl, srcs[i] + off, tl});
+    par_copy(ops);
+    HIP_TRY(hipMemcpyAsync(dbuf, hbuf, (size_t)n_src * tl,
+                           hipMemcpyHostToDevice, s.stream));
+    const MatmulCfg c = matmul_cfg((long)(tl >> 4), 1);
+    for (int g = 0; g < groups; g++) {
+      const int nj = std::min(4, n_out - 4 * g);
+      int rr = matmul_dispatch(s.stream, dbuf, dbuf, s.d_pparams + g, nj,
+                               n_src, false, c, tl, cps);
+      if (rr != ECX_OK) return rr;
+    }
+    HIP_TRY(hipMemcpyAsync(hbuf + (size_t)n_src * tl,
+                           dbuf + (size_t)n_src * tl, (size_t)n_out * tl,
+                           hipMemcpyDeviceToHost, s.stream));
+    HIP_TRY(hipEventRecord(s.ev_pipe[b], s.stream));
+    tl_of[b] = tl;
+    off_of[b] = off;
+  }
+  // drain the last one or two in-flight tiles
+  for (long t = std::max(0L, T - 2); t < T; t++) {
+    const int b = (int)(t & 1);
+    HIP_TRY(hipEventSynchronize(s.ev_pipe[b]));
+    uint8_t* hbuf = s.h_pipe + (size_t)b * cps * TS;
+    ops.clear();
+    for (int j = 0; j < n_out; j++)
+      if (outs[j])
+        ops.push_back({outs[j] + off_of[b],
+                       hbuf + ((size_t)n_src + j) * tl_of[b], tl_of[b]});
+    par_copy(ops);
+  }
+  return ECX_OK;
+}
+
 extern "C" {
 
 int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
@@ -1635,6 +1886,10 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
   const int si = (int)(ctx->rr++ % ctx->slots.size());
   Slot& s = ctx->slots[si];
   std::lock_guard<std::recursive_mutex> g(s.mu);
+  if (!ctx->is_w16() && !ctx->is_bitmatrix() && env_hostpipe())
+    return pipelined_matmul_host(ctx, s, data, k, parity, m,
+                                 ctx->gen.data() + (size_t)k * k,
+                                 chunk_bytes);
   int r = ensure_stage(ctx, s, (size_t)(k + m) * chunk_bytes);
   if (r != ECX_OK) return r;
   HIP_TRY(hipSetDevice(ctx->device));
@@ -1785,6 +2040,20 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
 
   Slot& s = ctx->slots[ctx->rr++ % ctx->slots.size()];
   std::lock_guard<std::recursive_mutex> g(s.mu);
+  if (env_hostpipe()) {
+    const uint8_t* srcs[ECX_MAX_K];
+    uint8_t* douts[ECX_MAX_K];
+    for (int i = 0; i < k; i++)
+      srcs[i] = chunks[plan.survivors[i]];  // NULL => zeros source
+                                            // (ErasureCodeIsa.cc:212-226)
+    const int ne = (int)plan.erased.size();
+    for (int j = 0; j < ne; j++) {
+      if (!chunks[plan.erased[j]]) return ECX_ERR_INVAL;
+      douts[j] = chunks[plan.erased[j]];
+    }
+    return pipelined_matmul_host(ctx, s, srcs, k, douts, ne,
+                                 plan.rows.data(), chunk_bytes);
+  }
   r = ensure_stage(ctx, s, (size_t)n * chunk_bytes);
   if (r != ECX_OK) return r;
   HIP_TRY(hipSetDevice(ctx->device));
