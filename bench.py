@@ -421,13 +421,33 @@ def bench_hostpath(args):
         ctx.encode_chunks(data)
         iters += 1
     dt = time.perf_counter() - t0
+    enc_gibs = iters * k * C / GIB / dt
+
+    # decode leg: m erasures (the worst single-call repair), host pointers
+    parity = ctx.encode_chunks(data)
+    chunks = [d.copy() for d in data] + [p.copy() for p in parity]
+    present = (1 << (k + m)) - 1
+    for e in range(m):  # erase first m data chunks
+        present &= ~(1 << e)
+    ctx.decode_chunks(chunks, present)  # warm
+    t0 = time.perf_counter()
+    dits = 0
+    while time.perf_counter() - t0 < 8.0:
+        ctx.decode_chunks(chunks, present)
+        dits += 1
+    ddt = time.perf_counter() - t0
+    for e in range(m):
+        assert np.array_equal(chunks[e], data[e]), "hostpath decode mismatch"
     print(json.dumps({
         "metric": "EC host-path (PCIe-inclusive) encode GiB/s",
-        "value": round(iters * k * C / GIB / dt, 3), "unit": "GiB/s",
+        "value": round(enc_gibs, 3), "unit": "GiB/s",
         "n_gpus": 1, "note": ("single-stripe ecx_encode_chunks_host incl. "
                               "H2D+D2H staging; drop-in plugin path, not "
                               "the device-resident metric"),
-        "config": {"k": k, "m": m, "chunk_bytes": C, "iters": iters},
+        "decode_gibs": round(dits * k * C / GIB / ddt, 3),
+        "decode_erasures": m,
+        "config": {"k": k, "m": m, "chunk_bytes": C, "iters": iters,
+                   "decode_iters": dits},
     }))
     ctx.close()
 
