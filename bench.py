@@ -426,9 +426,7 @@ def bench_hostpath(args):
     # decode leg: m erasures (the worst single-call repair), host pointers
     parity = ctx.encode_chunks(data)
     chunks = [d.copy() for d in data] + [p.copy() for p in parity]
-    present = (1 << (k + m)) - 1
-    for e in range(m):  # erase first m data chunks
-        present &= ~(1 << e)
+    present = [i >= m for i in range(k + m)]  # first m data chunks erased
     ctx.decode_chunks(chunks, present)  # warm
     t0 = time.perf_counter()
     dits = 0
