@@ -1704,10 +1704,12 @@ static int env_hostpipe() {
 static size_t env_hpipe_tile() {
   static const size_t v = [] {
     const char* e = getenv("ECX_HPIPE_TILE");
-    // 1 MiB default: the tile sweep (profiles/rocprof_r01_summary.md)
-    // showed throughput monotonically rising with tile size — per-tile
-    // event syncs cost more than cross-tile overlap buys
-    long x = e ? atol(e) : (1 << 20);
+    // 4 MiB default: the tile sweeps (profiles/rocprof_r01_summary.md)
+    // showed throughput monotonically rising with tile size at every
+    // chunk size tried — per-tile event syncs (~37 us each) cost more
+    // than cross-tile overlap buys, so prefer single-shot staging up to
+    // 4 MiB/chunk (pinned buffer: 2 x (k+m) x tile, grown on demand)
+    long x = e ? atol(e) : (4 << 20);
     if (x < (16 << 10)) x = 16 << 10;
     return (size_t)x & ~(size_t)15;
   }();
@@ -1843,19 +1845,17 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
                          hbuf + ((size_t)n_src + j) * tl_of[b], tl_of[b]});
       par_copy(ops);
     }
-    // gather tile t's sources packed at tl spacing, interleaving the CPU
-    // copy of source i+1 with the pinned DMA of source i (stream-ordered,
-    // so no events are needed; the tile sweep measured event syncs at
-    // ~37 us each, which ate the pipelining gain). Slots where srcs[i] is
-    // NULL are never read: cls==0 skips those sources in the kernel.
-    for (int i = 0; i < n_src; i++) {
-      if (!srcs[i]) continue;
-      ops.clear();
-      ops.push_back({hbuf + (size_t)i * tl, srcs[i] + off, tl});
-      par_copy(ops);
-      HIP_TRY(hipMemcpyAsync(dbuf + (size_t)i * tl, hbuf + (size_t)i * tl,
-                             tl, hipMemcpyHostToDevice, s.stream));
-    }
+    // gather tile t's sources packed at tl spacing (gaps where srcs[i] is
+    // NULL are never read: cls==0 skips those sources in the kernel), then
+    // ONE H2D DMA. Per-source interleaved copy+DMA was measured SLOWER
+    // (15.3 vs 19.0 GiB/s at 1 MiB chunks): the extra DMA submissions and
+    // OpenMP fork/joins cost more than the overlap hides.
+    ops.clear();
+    for (int i = 0; i < n_src; i++)
+      if (srcs[i]) ops.push_back({hbuf + (size_t)i * tl, srcs[i] + off, tl});
+    par_copy(ops);
+    HIP_TRY(hipMemcpyAsync(dbuf, hbuf, (size_t)n_src * tl,
+                           hipMemcpyHostToDevice, s.stream));
     const MatmulCfg c = matmul_cfg((long)(tl >> 4), 1);
     for (int g = 0; g < groups; g++) {
       const int nj = std::min(4, n_out - 4 * g);
