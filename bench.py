@@ -408,13 +408,46 @@ def bench_lrc(args):
 def bench_hostpath(args):
     """PCIe-inclusive plugin-path probe (single-stripe host-pointer calls,
     the drop-in path): reported separately from the device-resident metric
-    per DESIGN.md §5 — never the headline value."""
+    per DESIGN.md §5 — never the headline value. --threads N measures N
+    concurrent callers (the OSD's PG workers; ctypes releases the GIL, and
+    the context round-robins its stream slots --streams wide)."""
     import ceph_amd
     k, m, C = args.k, args.m, args.chunk_bytes
-    ctx = ceph_amd.EcContext(k, m, args.technique, device=0)
+    ctx = ceph_amd.EcContext(k, m, args.technique, device=0,
+                             n_streams=args.streams)
     rng = np.random.default_rng(args.seed)
     data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
     ctx.encode_chunks(data)  # warm
+
+    if args.threads > 1:
+        from concurrent.futures import ThreadPoolExecutor
+        datas = [[rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+                 for _ in range(args.threads)]
+
+        def worker(ti, stop):
+            n = 0
+            while time.perf_counter() < stop:
+                ctx.encode_chunks(datas[ti])
+                n += 1
+            return n
+        with ThreadPoolExecutor(args.threads) as ex:
+            stop = time.perf_counter() + 8.0
+            t0 = time.perf_counter()
+            counts = list(ex.map(lambda ti: worker(ti, stop),
+                                 range(args.threads)))
+        dt = time.perf_counter() - t0
+        print(json.dumps({
+            "metric": "EC host-path (PCIe-inclusive) encode GiB/s",
+            "value": round(sum(counts) * k * C / GIB / dt, 3),
+            "unit": "GiB/s", "n_gpus": 1,
+            "note": (f"{args.threads} concurrent host threads over "
+                     f"{args.streams} stream slots; drop-in plugin path"),
+            "config": {"k": k, "m": m, "chunk_bytes": C,
+                       "threads": args.threads, "streams": args.streams,
+                       "iters": sum(counts)},
+        }))
+        ctx.close()
+        return
     t0 = time.perf_counter()
     iters = 0
     while time.perf_counter() - t0 < 8.0:
@@ -466,6 +499,8 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--no-selfcheck", action="store_true")
     ap.add_argument("--streams", type=int, default=2)
+    ap.add_argument("--threads", type=int, default=1,
+                    help="concurrent host-path callers (hostpath preset)")
     ap.add_argument("--config", choices=["rs83", "cauchy104", "mixed",
                                          "hostpath", "slices", "lrc"],
                     default="rs83",
