@@ -1726,7 +1726,18 @@ static void par_copy(const std::vector<CopyOp>& ops) {
   size_t total = 0;
   for (const auto& o : ops) total += o.n;
 #ifdef _OPENMP
-  if (total >= (256 << 10)) {
+  // concurrency guard: each calling thread gets its own OpenMP team, and
+  // many teams of spinning workers interfere with each other and with the
+  // DMA-submit threads (8 concurrent plugin callers measured 8.6 GiB/s
+  // aggregate vs 17 for ONE). Bound the parallel teams; extra concurrent
+  // callers copy serially (29 GiB/s per core — aggregate still scales).
+  static std::atomic<int> active{0};
+  struct Scope {
+    std::atomic<int>& a;
+    ~Scope() { a.fetch_sub(1, std::memory_order_relaxed); }
+  } scope{active};
+  const int slot = active.fetch_add(1, std::memory_order_relaxed);
+  if (total >= (256 << 10) && slot < 2) {
     static const int nthr = [] {
       const char* e = getenv("ECX_HPIPE_THREADS");
       int x = e ? atoi(e) : 8;
