@@ -30,6 +30,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <dlfcn.h>
 #include <atomic>
 #include <cstdint>
 #include <algorithm>
@@ -1726,18 +1727,27 @@ static void par_copy(const std::vector<CopyOp>& ops) {
   size_t total = 0;
   for (const auto& o : ops) total += o.n;
 #ifdef _OPENMP
-  // concurrency guard: each calling thread gets its own OpenMP team, and
-  // many teams of spinning workers interfere with each other and with the
-  // DMA-submit threads (8 concurrent plugin callers measured 8.6 GiB/s
-  // aggregate vs 17 for ONE). Bound the parallel teams; extra concurrent
-  // callers copy serially (29 GiB/s per core — aggregate still scales).
+  // Concurrency guard, from the threaded sweeps
+  // (profiles/rocprof_r01_summary.md): each calling thread gets its own
+  // cached OpenMP team whose workers SPIN for KMP_BLOCKTIME (200 ms
+  // default) after every region — 8 concurrent plugin callers meant 64
+  // spinning cores fighting the DMA-submit threads and collapsed the
+  // aggregate to 8.7 GiB/s vs 39.8 with teams of 1. So: park idle
+  // workers immediately (blocktime 0), and only the first concurrent
+  // caller fans out — the rest copy serially (29 GiB/s per core, and the
+  // PCIe link is the shared bound anyway).
+  static std::once_flag bt_once;
+  std::call_once(bt_once, [] {
+    if (void* f = dlsym(RTLD_DEFAULT, "kmp_set_blocktime"))
+      ((void (*)(int))f)(0);
+  });
   static std::atomic<int> active{0};
   struct Scope {
     std::atomic<int>& a;
     ~Scope() { a.fetch_sub(1, std::memory_order_relaxed); }
   } scope{active};
   const int slot = active.fetch_add(1, std::memory_order_relaxed);
-  if (total >= (256 << 10) && slot < 2) {
+  if (total >= (256 << 10) && slot < 1) {
     static const int nthr = [] {
       const char* e = getenv("ECX_HPIPE_THREADS");
       int x = e ? atoi(e) : 8;
