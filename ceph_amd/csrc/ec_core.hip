@@ -1803,6 +1803,68 @@ static int ensure_pparams(ecx_ctx* ctx, Slot& s) {
   return ECX_OK;
 }
 
+static size_t env_hpipe_max() {
+  static const size_t v = [] {
+    const char* e = getenv("ECX_HPIPE_MAX");
+    long x = e ? atol(e) : (256L << 20);
+    return (size_t)(x < (1 << 20) ? (1 << 20) : x);
+  }();
+  return v;
+}
+
+// Single-shot pinned staging for the legacy-launcher host paths (w16 and
+// bitmatrix techniques): gather chunks into the pinned pipe buffer at
+// their slot offsets (NULL source => zeros chunk), ONE H2D DMA over the
+// source span, run the technique's launcher on the device pipe buffer,
+// per-output D2H, scatter. Same measured rationale as
+// pipelined_matmul_host: per-chunk pageable copies pay ~50 us each.
+// Caller holds the slot lock and bounds the size by env_hpipe_max().
+template <class LaunchFn>
+static int staged_host_call(ecx_ctx* ctx, Slot& s,
+                            const uint8_t* const* hsrc, const int* src_slot,
+                            int n_src, uint8_t* const* hout,
+                            const int* out_slot, int n_out, int n_slots,
+                            size_t chunk_bytes, LaunchFn&& launch) {
+  int r = ensure_pipe(ctx, s, (size_t)n_slots * chunk_bytes);
+  if (r != ECX_OK) return r;
+  HIP_TRY(hipSetDevice(ctx->device));
+  HIP_TRY(hipEventSynchronize(s.ev_pipe[0]));
+  HIP_TRY(hipEventSynchronize(s.ev_pipe[1]));
+  std::vector<CopyOp> ops;
+  int lo = n_slots, hi = 0;
+  for (int i = 0; i < n_src; i++) {
+    const int sl = src_slot[i];
+    lo = std::min(lo, sl);
+    hi = std::max(hi, sl + 1);
+    if (hsrc[i])
+      ops.push_back({s.h_pipe + (size_t)sl * chunk_bytes, hsrc[i],
+                     chunk_bytes});
+    else
+      std::memset(s.h_pipe + (size_t)sl * chunk_bytes, 0, chunk_bytes);
+  }
+  par_copy(ops);
+  HIP_TRY(hipMemcpyAsync(s.d_pipe + (size_t)lo * chunk_bytes,
+                         s.h_pipe + (size_t)lo * chunk_bytes,
+                         (size_t)(hi - lo) * chunk_bytes,
+                         hipMemcpyHostToDevice, s.stream));
+  r = launch(s.d_pipe);
+  if (r != ECX_OK) return r;
+  for (int j = 0; j < n_out; j++) {
+    if (!hout[j]) continue;
+    HIP_TRY(hipMemcpyAsync(s.h_pipe + (size_t)out_slot[j] * chunk_bytes,
+                           s.d_pipe + (size_t)out_slot[j] * chunk_bytes,
+                           chunk_bytes, hipMemcpyDeviceToHost, s.stream));
+  }
+  HIP_TRY(hipStreamSynchronize(s.stream));
+  ops.clear();
+  for (int j = 0; j < n_out; j++)
+    if (hout[j])
+      ops.push_back({hout[j], s.h_pipe + (size_t)out_slot[j] * chunk_bytes,
+                     chunk_bytes});
+  par_copy(ops);
+  return ECX_OK;
+}
+
 // Generic host-pointer matmul (w=8 table techniques): srcs[i] == NULL is
 // the zeros-chunk convention (skipped via cls), outs[j] == NULL skips the
 // scatter of that output. Caller holds the slot lock.
@@ -1919,6 +1981,30 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
     return pipelined_matmul_host(ctx, s, data, k, parity, m,
                                  ctx->gen.data() + (size_t)k * k,
                                  chunk_bytes);
+  if (env_hostpipe() && (size_t)(k + m) * chunk_bytes <= env_hpipe_max()) {
+    // w16 / bitmatrix: single-shot pinned staging around the technique's
+    // own launcher (they keep per-call param staging, so no tiling)
+    int src_ids[ECX_MAX_K], out_ids[ECX_MAX_K];
+    for (int i = 0; i < k; i++) src_ids[i] = i;
+    for (int j = 0; j < m; j++) out_ids[j] = k + j;
+    return staged_host_call(
+        ctx, s, data, src_ids, k, parity, out_ids, m, k + m, chunk_bytes,
+        [&](uint8_t* d) -> int {
+          if (ctx->is_w16())
+            return run_matmul16(ctx, si, d, d, src_ids, k, out_ids, m,
+                                ctx->gen16.data() + (size_t)k * k, 1,
+                                chunk_bytes, false, false);
+          for (int j0 = 0; j0 < m; j0 += ECX_MAX_OUT) {
+            int nj = std::min(ECX_MAX_OUT, m - j0);
+            int rr = run_bitmatrix(
+                ctx, si, d, d, src_ids, k, out_ids + j0, nj,
+                ctx->bitmat.data() + (size_t)(j0 * ctx->w) * k * ctx->w, 1,
+                chunk_bytes, false);
+            if (rr != ECX_OK) return rr;
+          }
+          return ECX_OK;
+        });
+  }
   int r = ensure_stage(ctx, s, (size_t)(k + m) * chunk_bytes);
   if (r != ECX_OK) return r;
   HIP_TRY(hipSetDevice(ctx->device));
@@ -1995,6 +2081,23 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
     const int si = (int)(ctx->rr++ % ctx->slots.size());
     Slot& s = ctx->slots[si];
     std::lock_guard<std::recursive_mutex> g(s.mu);
+    if (env_hostpipe() && (size_t)n * chunk_bytes <= env_hpipe_max()) {
+      const int ne = (int)plan.erased.size();
+      const uint8_t* srcs[ECX_MAX_K];
+      uint8_t* douts[ECX_MAX_K];
+      for (int i = 0; i < k; i++) srcs[i] = chunks[plan.survivors[i]];
+      for (int j = 0; j < ne; j++) {
+        if (!chunks[plan.erased[j]]) return ECX_ERR_INVAL;
+        douts[j] = chunks[plan.erased[j]];
+      }
+      return staged_host_call(
+          ctx, s, srcs, plan.survivors.data(), k, douts,
+          plan.erased.data(), ne, n, chunk_bytes, [&](uint8_t* d) {
+            return run_matmul16(ctx, si, d, d, plan.survivors.data(), k,
+                                plan.erased.data(), ne, plan.rows.data(),
+                                1, chunk_bytes, false, false);
+          });
+    }
     r = ensure_stage(ctx, s, (size_t)n * chunk_bytes);
     if (r != ECX_OK) return r;
     HIP_TRY(hipSetDevice(ctx->device));
@@ -2030,6 +2133,30 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
     const int si = (int)(ctx->rr++ % ctx->slots.size());
     Slot& s = ctx->slots[si];
     std::lock_guard<std::recursive_mutex> g(s.mu);
+    if (env_hostpipe() && (size_t)n * chunk_bytes <= env_hpipe_max()) {
+      const int ne = (int)plan.erased.size();
+      const uint8_t* srcs[ECX_MAX_K];
+      uint8_t* douts[ECX_MAX_K];
+      for (int i = 0; i < k; i++) srcs[i] = chunks[plan.survivors[i]];
+      for (int j = 0; j < ne; j++) {
+        if (!chunks[plan.erased[j]]) return ECX_ERR_INVAL;
+        douts[j] = chunks[plan.erased[j]];
+      }
+      return staged_host_call(
+          ctx, s, srcs, plan.survivors.data(), k, douts,
+          plan.erased.data(), ne, n, chunk_bytes, [&](uint8_t* d) -> int {
+            for (int j0 = 0; j0 < ne; j0 += ECX_MAX_OUT) {
+              int nj = std::min(ECX_MAX_OUT, ne - j0);
+              int rr = run_bitmatrix(
+                  ctx, si, d, d, plan.survivors.data(), k,
+                  plan.erased.data() + j0, nj,
+                  plan.rows.data() + (size_t)j0 * ctx->w * k * ctx->w, 1,
+                  chunk_bytes, false);
+              if (rr != ECX_OK) return rr;
+            }
+            return ECX_OK;
+          });
+    }
     r = ensure_stage(ctx, s, (size_t)n * chunk_bytes);
     if (r != ECX_OK) return r;
     HIP_TRY(hipSetDevice(ctx->device));
