@@ -1,9 +1,7 @@
 # ec-mi355x — build everything (CPU oracle, GPU core, plugin harness)
 HIPCC    ?= hipcc
 GPU_ARCH ?= gfx950
-# -fopenmp: host-side parallel gather/scatter in the pipelined
-# host-pointer path (pipelined_matmul_host); device pass unaffected
-HIPFLAGS ?= --offload-arch=$(GPU_ARCH) -O3 -std=c++17 -fPIC -fopenmp
+HIPFLAGS ?= --offload-arch=$(GPU_ARCH) -O3 -std=c++17 -fPIC
 CXX      ?= g++
 CXXFLAGS ?= -O2 -std=c++17 -fPIC -Wall
 HARNESS   = ceph_amd/harness

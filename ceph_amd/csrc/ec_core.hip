@@ -30,13 +30,16 @@
 
 #include <hip/hip_runtime.h>
 
-#include <dlfcn.h>
+
 #include <atomic>
 #include <cstdint>
 #include <algorithm>
 #include <cstring>
 #include <list>
 #include <map>
+#include <chrono>
+#include <condition_variable>
+#include <thread>
 #include <mutex>
 #include <string>
 #include <vector>
@@ -1723,50 +1726,139 @@ struct CopyOp {
   size_t n;
 };
 
+// Shared copy pool for the pinned-staging gather/scatter. Per-caller
+// OpenMP teams were measured and rejected (profiles summary): with the
+// default spin-wait, 8 concurrent plugin callers meant 64 spinning
+// workers fighting the DMA-submit threads (8.7 GiB/s aggregate vs 39.8
+// serial); with parked workers (blocktime 0) the single caller paid 8
+// futex wakeups per region (11.3 vs 17 GiB/s). One process-wide pool
+// avoids both: helpers stay hot while ANY caller has work (no wakeups in
+// steady state, no idle spin against the HIP runtime), and callers drain
+// the queue themselves, so the serial memcpy rate is the floor.
+class CopyPool {
+ public:
+  static CopyPool& inst() {
+    static CopyPool p;
+    return p;
+  }
+
+  void run(const std::vector<CopyOp>& ops) {
+    constexpr size_t PIECE = 512 << 10;
+    size_t n_items = 0;
+    for (const auto& o : ops) n_items += (o.n + PIECE - 1) / PIECE;
+    std::atomic<size_t> remaining{n_items};
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      for (const auto& o : ops)
+        for (size_t off = 0; off < o.n; off += PIECE)
+          q_.push_back({o.dst + off, o.src + off,
+                        std::min(PIECE, o.n - off), &remaining});
+      pending_.fetch_add(n_items, std::memory_order_release);
+    }
+    cv_.notify_all();
+    // help drain (possibly other callers' pieces — work conservation)
+    for (;;) {
+      Item it;
+      {
+        std::lock_guard<std::mutex> lk(mu_);
+        if (!pop(it)) break;
+      }
+      do_copy(it);
+    }
+    // our pieces may still be in helpers' hands
+    while (remaining.load(std::memory_order_acquire) != 0) relax();
+  }
+
+ private:
+  struct Item {
+    uint8_t* dst;
+    const uint8_t* src;
+    size_t n;
+    std::atomic<size_t>* done;
+  };
+
+  CopyPool() {
+    const char* e = getenv("ECX_HPIPE_THREADS");
+    int x = e ? atoi(e) : 7;  // helpers; the caller is the +1
+    n_workers_ = x < 0 ? 0 : (x > 63 ? 63 : x);
+    for (int i = 0; i < n_workers_; i++)
+      workers_.emplace_back([this] { loop(); });
+  }
+
+  ~CopyPool() {
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      stop_ = true;
+    }
+    cv_.notify_all();
+    for (auto& w : workers_) w.join();
+  }
+
+  static void relax() {
+#if defined(__x86_64__)
+    __builtin_ia32_pause();
+#else
+    std::this_thread::yield();
+#endif
+  }
+
+  bool pop(Item& it) {
+    if (qhead_ >= q_.size()) return false;
+    it = q_[qhead_++];
+    pending_.fetch_sub(1, std::memory_order_relaxed);
+    if (qhead_ == q_.size()) {
+      q_.clear();
+      qhead_ = 0;
+    }
+    return true;
+  }
+
+  static void do_copy(const Item& it) {
+    std::memcpy(it.dst, it.src, it.n);
+    it.done->fetch_sub(1, std::memory_order_acq_rel);
+  }
+
+  void loop() {
+    std::unique_lock<std::mutex> lk(mu_);
+    for (;;) {
+      if (stop_) return;
+      Item it;
+      if (pop(it)) {
+        lk.unlock();
+        do_copy(it);
+        lk.lock();
+        continue;
+      }
+      // brief unlocked spin: back-to-back plugin calls arrive every few
+      // tens of us, so staying hot beats a park/wake round trip
+      lk.unlock();
+      int spins = 20000;
+      while (--spins > 0 && pending_.load(std::memory_order_relaxed) == 0)
+        relax();
+      lk.lock();
+      if (pending_.load(std::memory_order_relaxed) == 0 && !stop_)
+        cv_.wait_for(lk, std::chrono::milliseconds(50));
+    }
+  }
+
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::vector<Item> q_;
+  size_t qhead_ = 0;
+  std::atomic<size_t> pending_{0};
+  bool stop_ = false;
+  int n_workers_ = 0;
+  std::vector<std::thread> workers_;
+};
+
 static void par_copy(const std::vector<CopyOp>& ops) {
   size_t total = 0;
   for (const auto& o : ops) total += o.n;
-#ifdef _OPENMP
-  // Concurrency guard, from the threaded sweeps
-  // (profiles/rocprof_r01_summary.md): each calling thread gets its own
-  // cached OpenMP team whose workers SPIN for KMP_BLOCKTIME (200 ms
-  // default) after every region — 8 concurrent plugin callers meant 64
-  // spinning cores fighting the DMA-submit threads and collapsed the
-  // aggregate to 8.7 GiB/s vs 39.8 with teams of 1. So: park idle
-  // workers immediately (blocktime 0), and only the first concurrent
-  // caller fans out — the rest copy serially (29 GiB/s per core, and the
-  // PCIe link is the shared bound anyway).
-  static std::once_flag bt_once;
-  std::call_once(bt_once, [] {
-    if (void* f = dlsym(RTLD_DEFAULT, "kmp_set_blocktime"))
-      ((void (*)(int))f)(0);
-  });
-  static std::atomic<int> active{0};
-  struct Scope {
-    std::atomic<int>& a;
-    ~Scope() { a.fetch_sub(1, std::memory_order_relaxed); }
-  } scope{active};
-  const int slot = active.fetch_add(1, std::memory_order_relaxed);
-  if (total >= (256 << 10) && slot < 1) {
-    static const int nthr = [] {
-      const char* e = getenv("ECX_HPIPE_THREADS");
-      int x = e ? atoi(e) : 8;
-      return x < 1 ? 1 : (x > 64 ? 64 : x);
-    }();
-    constexpr size_t PIECE = 256 << 10;
-    std::vector<CopyOp> pieces;
-    pieces.reserve(ops.size() * 4);
-    for (const auto& o : ops)
-      for (size_t off = 0; off < o.n; off += PIECE)
-        pieces.push_back({o.dst + off, o.src + off,
-                          std::min(PIECE, o.n - off)});
-#pragma omp parallel for num_threads(nthr) schedule(static)
-    for (long i = 0; i < (long)pieces.size(); i++)
-      std::memcpy(pieces[i].dst, pieces[i].src, pieces[i].n);
+  if (total < (256 << 10)) {
+    for (const auto& o : ops) std::memcpy(o.dst, o.src, o.n);
     return;
   }
-#endif
-  for (const auto& o : ops) std::memcpy(o.dst, o.src, o.n);
+  CopyPool::inst().run(ops);
 }
 
 static int ensure_pipe(ecx_ctx* ctx, Slot& s, size_t bytes) {
