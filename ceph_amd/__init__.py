@@ -286,18 +286,19 @@ class EcContext:
         return ms.value
 
     # ---- host-pointer (plugin) path ----
-    def encode_chunks(self, data, chunk_bytes=None):
+    def encode_chunks(self, data, chunk_bytes=None, out=None):
         """data: list of k uint8 arrays (None => zeros chunk). Returns m
         parity arrays. Mirrors encode_chunks marshalling
         (ErasureCodeJerasure.cc:121-164)."""
         sizes = {d.nbytes for d in data if d is not None}
         assert len(sizes) == 1
         n = sizes.pop()
-        parity = [np.zeros(n, dtype=np.uint8) for _ in range(self.m)]
+        if out is None:
+            out = [np.zeros(n, dtype=np.uint8) for _ in range(self.m)]
         _ck(lib().ecx_encode_chunks_host(self._h, _ptr_array(data),
-                                         _ptr_array(parity), n),
+                                         _ptr_array(out), n),
             "ecx_encode_chunks_host")
-        return parity
+        return out
 
     def decode_chunks(self, chunks, present):
         """chunks: list of k+m uint8 arrays; erased entries (present[i]

@@ -1743,7 +1743,7 @@ class CopyPool {
   }
 
   void run(const std::vector<CopyOp>& ops) {
-    constexpr size_t PIECE = 512 << 10;
+    constexpr size_t PIECE = 1 << 20;
     size_t n_items = 0;
     for (const auto& o : ops) n_items += (o.n + PIECE - 1) / PIECE;
     std::atomic<size_t> remaining{n_items};
@@ -1755,7 +1755,10 @@ class CopyPool {
                         std::min(PIECE, o.n - off), &remaining});
       pending_.fetch_add(n_items, std::memory_order_release);
     }
-    cv_.notify_all();
+    // spinning helpers see pending_ without the lock; wake only the ones
+    // that actually parked (notify_all costs the CALLER a futex syscall
+    // per parked helper — measured as a single-caller regression)
+    if (parked_.load(std::memory_order_acquire) > 0) cv_.notify_all();
     // help drain (possibly other callers' pieces — work conservation)
     for (;;) {
       Item it;
@@ -1829,15 +1832,19 @@ class CopyPool {
         lk.lock();
         continue;
       }
-      // brief unlocked spin: back-to-back plugin calls arrive every few
-      // tens of us, so staying hot beats a park/wake round trip
+      // unlocked spin ~2 ms: single-caller stripes arrive ~1 ms apart,
+      // so staying hot through the gap beats a park/wake round trip; an
+      // idle context parks all helpers after the window
       lk.unlock();
-      int spins = 20000;
+      int spins = 60000;
       while (--spins > 0 && pending_.load(std::memory_order_relaxed) == 0)
         relax();
       lk.lock();
-      if (pending_.load(std::memory_order_relaxed) == 0 && !stop_)
+      if (pending_.load(std::memory_order_relaxed) == 0 && !stop_) {
+        parked_.fetch_add(1, std::memory_order_release);
         cv_.wait_for(lk, std::chrono::milliseconds(50));
+        parked_.fetch_sub(1, std::memory_order_release);
+      }
     }
   }
 
@@ -1846,6 +1853,7 @@ class CopyPool {
   std::vector<Item> q_;
   size_t qhead_ = 0;
   std::atomic<size_t> pending_{0};
+  std::atomic<int> parked_{0};
   bool stop_ = false;
   int n_workers_ = 0;
   std::vector<std::thread> workers_;
