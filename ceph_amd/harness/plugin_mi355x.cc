@@ -50,16 +50,31 @@ class ErasureCodeMi355x final : public ErasureCode {
   size_t get_minimum_granularity() override { return 16; }
 
   plugin_flags get_supported_optimizations() const override {
-    // the subset the conformance tests verify (cf. ErasureCodeIsa.h:68-79;
-    // CRC composition is not implemented yet, so not claimed); parity-delta
-    // is matrix-technique-only (cauchy_orig delta schedules are a later
-    // round — claim only what works, the conformance rule)
+    // mirror the upstream plugin that owns each technique, flag for flag:
+    // isa techniques (ErasureCodeIsa.h:66-79) claim OPTIMIZED always and
+    // CRC for reed_sol_van (cauchy only at m=1); jerasure techniques
+    // (ErasureCodeJerasure.h:52-63) claim OPTIMIZED only for reed_sol_van
+    // and CRC for everything except reed_sol_van and cauchy_orig. The CRC
+    // flag is a capability declaration the OSD's scrub consumes — there
+    // is no plugin-side CRC method to implement. Parity-delta is
+    // matrix-technique-only here (cauchy_orig delta schedules are a later
+    // round — claim only what works, the conformance rule).
     plugin_flags f = FLAG_EC_PLUGIN_PARTIAL_READ_OPTIMIZATION |
                      FLAG_EC_PLUGIN_PARTIAL_WRITE_OPTIMIZATION |
                      FLAG_EC_PLUGIN_ZERO_INPUT_ZERO_OUTPUT_OPTIMIZATION |
-                     FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED |
                      FLAG_EC_PLUGIN_DIRECT_READS;
     if (!is_bitmatrix()) f |= FLAG_EC_PLUGIN_PARITY_DELTA_OPTIMIZATION;
+    if (technique_ == "reed_sol_van" || technique_ == "cauchy") {
+      // isa-matrix techniques
+      f |= FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED;
+      if (technique_ == "reed_sol_van" ||
+          (technique_ == "cauchy" && m_ == 1))
+        f |= FLAG_EC_PLUGIN_CRC_ENCODE_DECODE_SUPPORT;
+    } else if (technique_ == "jerasure_reed_sol_van") {
+      f |= FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED;  // jerasure reed_sol_van
+    } else if (!is_bitmatrix()) {
+      f |= FLAG_EC_PLUGIN_CRC_ENCODE_DECODE_SUPPORT;  // e.g. reed_sol_r6_op
+    }
     return f;
   }
 

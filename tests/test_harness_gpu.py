@@ -115,3 +115,33 @@ def test_lrc_decode_exhaustive_gpu():
                   "-s", str(4 * 65536), "-i", "2", "-w", "decode",
                   "-e", "1", "-E", "exhaustive")
     assert r.returncode == 0, r.stderr + r.stdout
+
+
+@pytest.mark.parametrize("profile,expected", [
+    (["-P", "technique=reed_sol_van", "-P", "k=8", "-P", "m=3"],
+     "partialread,partialwrite,zeroinout,paritydelta,optimizedsupport,"
+     "crcencodedecode,directreads"),
+    (["-P", "technique=cauchy", "-P", "k=4", "-P", "m=1"],
+     "partialread,partialwrite,zeroinout,paritydelta,optimizedsupport,"
+     "crcencodedecode,directreads"),
+    (["-P", "technique=cauchy", "-P", "k=4", "-P", "m=2"],
+     "partialread,partialwrite,zeroinout,paritydelta,optimizedsupport,"
+     "directreads"),
+    (["-P", "technique=jerasure_reed_sol_van", "-P", "k=4", "-P", "m=2"],
+     "partialread,partialwrite,zeroinout,paritydelta,optimizedsupport,"
+     "directreads"),
+    (["-P", "technique=reed_sol_r6_op", "-P", "k=4", "-P", "m=2"],
+     "partialread,partialwrite,zeroinout,paritydelta,crcencodedecode,"
+     "directreads"),
+    (["-P", "technique=cauchy_orig", "-P", "k=4", "-P", "m=2"],
+     "partialread,partialwrite,zeroinout,directreads"),
+])
+def test_mi355x_flags_mirror_reference(profile, expected):
+    """Per-technique optimization flags mirror the owning reference
+    plugin exactly (ErasureCodeIsa.h:66-79, ErasureCodeJerasure.h:52-63):
+    isa techniques claim OPTIMIZED always and CRC for reed_sol_van /
+    cauchy-at-m=1; jerasure techniques claim OPTIMIZED only for
+    reed_sol_van and CRC for all but reed_sol_van and cauchy_orig."""
+    r = run_bench("-p", "mi355x", *profile, "--flags")
+    assert r.returncode == 0, r.stderr
+    assert r.stdout.strip() == expected
