@@ -1957,7 +1957,10 @@ static int staged_host_call(ecx_ctx* ctx, Slot& s,
                          (size_t)(hi - lo) * chunk_bytes,
                          hipMemcpyHostToDevice, s.stream));
   r = launch(s.d_pipe);
-  if (r != ECX_OK) return r;
+  if (r != ECX_OK) {
+    (void)hipStreamSynchronize(s.stream);  // quiesce before buffer reuse
+    return r;
+  }
   for (int j = 0; j < n_out; j++) {
     if (!hout[j]) continue;
     HIP_TRY(hipMemcpyAsync(s.h_pipe + (size_t)out_slot[j] * chunk_bytes,
@@ -2025,6 +2028,9 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
     HIP_TRY(hipMemcpyAsync(s.d_pparams, s.h_pparams,
                            (size_t)groups * sizeof(EcLaunchParams),
                            hipMemcpyHostToDevice, s.stream));
+    // cover the in-flight upload so the next call's safety syncs see it
+    // even if this call errors out before the first tile's record
+    HIP_TRY(hipEventRecord(s.ev_pipe[0], s.stream));
     s.pparams_kind = want_kind;
   }
 
@@ -2065,7 +2071,10 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
       const int nj = std::min(4, n_out - 4 * g);
       int rr = matmul_dispatch(s.stream, dbuf, dbuf, s.d_pparams + g, nj,
                                n_src, false, c, tl, cps);
-      if (rr != ECX_OK) return rr;
+      if (rr != ECX_OK) {
+        (void)hipStreamSynchronize(s.stream);  // quiesce in-flight tiles
+        return rr;
+      }
     }
     HIP_TRY(hipMemcpyAsync(hbuf + (size_t)n_src * tl,
                            dbuf + (size_t)n_src * tl, (size_t)n_out * tl,
