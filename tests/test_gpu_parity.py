@@ -413,3 +413,46 @@ def test_cauchy_orig_packetsize_variants(pkt):
             assert np.array_equal(got[j], want[j]), (pkt, j)
     finally:
         ctx.close()
+
+
+def test_multi_context_concurrent_techniques():
+    """An OSD hosts many EC profiles at once: several live contexts of
+    different techniques driven from concurrent threads must not
+    interfere (separate slot pools, shared device)."""
+    import threading
+
+    import ceph_amd
+    import oracle
+
+    shapes = [("reed_sol_van", 8, 3), ("cauchy", 6, 2),
+              ("jerasure_reed_sol_van", 4, 2), ("reed_sol_van", 10, 4)]
+    C = 256 * 1024
+    ctxs = [ceph_amd.EcContext(k, m, t, device=0) for (t, k, m) in shapes]
+    errs = []
+
+    def worker(idx):
+        t, k, m = shapes[idx]
+        rng = np.random.default_rng(idx)
+        try:
+            for _ in range(8):
+                data = [rng.integers(0, 256, C, dtype=np.uint8)
+                        for _ in range(k)]
+                got = ctxs[idx].encode_chunks(data)
+                base = "reed_sol_van" if t == "reed_sol_van" else t
+                want = oracle.encode(base, k, m, data)
+                for j in range(m):
+                    if not np.array_equal(got[j], want[j]):
+                        errs.append((idx, j))
+                        return
+        except Exception as e:  # noqa: BLE001 - surface into main thread
+            errs.append((idx, repr(e)))
+
+    threads = [threading.Thread(target=worker, args=(i,))
+               for i in range(len(shapes))]
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join()
+    for c in ctxs:
+        c.close()
+    assert not errs, errs
