@@ -22,7 +22,7 @@ ceph_amd/libec_mi355x_core.so: ceph_amd/csrc/ec_core.hip ceph_amd/csrc/gf.cpp ce
 #      dlopen, like real Ceph plugins do — hence -rdynamic on binaries) ----
 HARNESS_HDRS = $(HARNESS)/ec_types.h $(HARNESS)/erasure_code.h $(HARNESS)/erasure_code_plugin.h
 
-harness: $(HARNESS)/ec_benchmark $(HARNESS)/registry_selftest \
+harness: $(HARNESS)/ec_benchmark $(HARNESS)/registry_selftest $(HARNESS)/ec_non_regression \
          $(HARNESS)/libec_mi355x.so $(HARNESS)/libec_oracle.so $(HARNESS)/libec_lrc.so $(HARNESS)/libec_shec.so $(HARNESS)/libec_clay.so \
          $(HARNESS)/libec_fix_missing_version.so $(HARNESS)/libec_fix_bad_version.so \
          $(HARNESS)/libec_fix_missing_init.so $(HARNESS)/libec_fix_fail_init.so \
@@ -33,6 +33,10 @@ $(HARNESS)/ec_benchmark: $(HARNESS)/ec_benchmark.cc $(HARNESS)/erasure_code.cc $
 
 $(HARNESS)/registry_selftest: $(HARNESS)/registry_selftest.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc $(HARNESS_HDRS)
 	$(CXX) $(CXXFLAGS) -rdynamic $(HARNESS)/registry_selftest.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc -ldl -o $@
+
+# non-regression corpus tool (mirror of ceph_erasure_code_non_regression)
+$(HARNESS)/ec_non_regression: $(HARNESS)/ec_non_regression.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc $(HARNESS_HDRS)
+	$(CXX) $(CXXFLAGS) -rdynamic $(HARNESS)/ec_non_regression.cc $(HARNESS)/erasure_code.cc $(HARNESS)/erasure_code_plugin.cc -ldl -o $@
 
 # product plugin: links the HIP core; undefined harness symbols resolve from
 # the loading binary (RTLD_NOW), mirroring real Ceph plugin linkage
@@ -75,6 +79,6 @@ $(HARNESS)/libec_fix_no_register.so: $(HARNESS)/plugin_fixture.cc $(HARNESS_HDRS
 
 clean:
 	$(MAKE) -C oracle clean
-	rm -f ceph_amd/libec_mi355x_core.so $(HARNESS)/*.so $(HARNESS)/ec_benchmark $(HARNESS)/registry_selftest
+	rm -f ceph_amd/libec_mi355x_core.so $(HARNESS)/*.so $(HARNESS)/ec_benchmark $(HARNESS)/registry_selftest $(HARNESS)/ec_non_regression
 
 .PHONY: all oracle core harness clean
