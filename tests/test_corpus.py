@@ -1,0 +1,55 @@
+"""Committed non-regression corpus (tests/golden/corpus): chunk archives
+generated once by the CPU oracle plugin via ec_non_regression, pinning
+bit-exactness across future versions of this repo — the role the
+ceph-erasure-code-corpus replay plays for the reference.
+Every config is re-checked by the oracle plugin on CPU and by the mi355x
+plugin on GPU."""
+import os
+import shutil
+import subprocess
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+HARNESS = os.path.join(ROOT, "ceph_amd", "harness")
+CORPUS = os.path.join(ROOT, "tests", "golden", "corpus")
+TOOL = os.path.join(HARNESS, "ec_non_regression")
+
+CONFIGS = sorted(d for d in os.listdir(CORPUS)
+                 if d.startswith("plugin=oracle "))
+
+
+def args_of(dirname):
+    """Recover the tool invocation from the directory name (the layout
+    embeds stripe width and profile params in order)."""
+    args = []
+    for tok in dirname.split(" ")[1:]:
+        k, v = tok.split("=", 1)
+        if k == "stripe-width":
+            args += ["-s", v]
+        else:
+            args += ["-P", tok]
+    return args
+
+
+def run_check(base, plugin, dirname):
+    if not os.path.exists(TOOL):
+        pytest.skip("harness not built")
+    return subprocess.run(
+        [TOOL, "-d", HARNESS, "--base", str(base), "-p", plugin,
+         *args_of(dirname), "--check"], capture_output=True, text=True)
+
+
+@pytest.mark.parametrize("dirname", CONFIGS)
+def test_oracle_matches_committed_corpus(dirname):
+    r = run_check(CORPUS, "oracle", dirname)
+    assert r.returncode == 0, (dirname, r.stderr)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dirname", CONFIGS)
+def test_gpu_plugin_matches_committed_corpus(dirname, tmp_path):
+    renamed = dirname.replace("plugin=oracle", "plugin=mi355x", 1)
+    shutil.copytree(os.path.join(CORPUS, dirname), tmp_path / renamed)
+    r = run_check(tmp_path, "mi355x", renamed)
+    assert r.returncode == 0, (dirname, r.stderr)
