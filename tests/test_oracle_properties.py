@@ -243,3 +243,22 @@ class TestBitmatrixCauchyOrig:
         data = [np.zeros(100, np.uint8) for _ in range(k)]
         with pytest.raises(ValueError):
             oracle.bitmatrix_encode(k, m, data, p)
+
+
+@pytest.mark.parametrize("tech", ["reed_sol_van", "cauchy",
+                                  "jerasure_reed_sol_van"])
+def test_byte_column_locality(tech):
+    """The PartialWrite optimization's underlying property
+    (TestErasureCodePlugins.cc:173-257): parity byte b depends only on
+    data bytes at offset b, so flipping one data byte changes exactly
+    that column of every parity chunk."""
+    k, m, C = 5, 3, 256
+    rng = np.random.default_rng(11)
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    base = oracle.encode(tech, k, m, data)
+    data2 = [d.copy() for d in data]
+    data2[2][97] ^= 0x5A
+    mod = oracle.encode(tech, k, m, data2)
+    for j in range(m):
+        diff = np.flatnonzero(base[j] != mod[j])
+        assert diff.tolist() == [97], (tech, j, diff)
