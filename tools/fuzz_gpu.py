@@ -20,13 +20,13 @@ def main():
     rng = np.random.default_rng(args.seed)
     stop = time.monotonic() + args.seconds
     cases = 0
+    # reed_sol_r6_op is a harness-level alias of the isa RS-van matrix at
+    # m=2 (tested in test_harness*); the C-ABI exposes the 5 base ids
     techs = ["reed_sol_van", "cauchy", "jerasure_reed_sol_van",
-             "reed_sol_r6_op", "jerasure_reed_sol_van_w16", "cauchy_orig"]
+             "jerasure_reed_sol_van_w16", "cauchy_orig"]
     while time.monotonic() < stop:
         tech = techs[rng.integers(0, len(techs))]
-        if tech == "reed_sol_r6_op":
-            k, m = int(rng.integers(2, 17)), 2
-        elif tech == "cauchy_orig":
+        if tech == "cauchy_orig":
             k, m = int(rng.integers(2, 13)), int(rng.integers(1, 5))
         else:
             k, m = int(rng.integers(2, 21)), int(rng.integers(1, 5))
@@ -48,15 +48,12 @@ def main():
                     rng.integers(0, 256, C, dtype=np.uint8)
                     for _ in range(k)]
             full = [np.zeros(C, np.uint8) if d is None else d for d in data]
-            base = "reed_sol_van" if tech == "reed_sol_r6_op" else tech
             if tech == "cauchy_orig":
                 want = oracle.bitmatrix_encode(k, m, full, pkt)
             elif tech == "jerasure_reed_sol_van_w16":
                 want = oracle.encode_w16(k, m, full)
-            elif tech == "reed_sol_r6_op":
-                want = oracle.encode("reed_sol_van", k, m, full)
             else:
-                want = oracle.encode(base, k, m, full)
+                want = oracle.encode(tech, k, m, full)
             got = ctx.encode_chunks(data)
             for j in range(m):
                 assert np.array_equal(got[j], want[j]), ("enc", repro, j)
