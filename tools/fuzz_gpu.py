@@ -5,7 +5,12 @@ prints the full reproducer tuple. Exit 0 = clean."""
 import argparse
 import time
 
+import os
+import sys
+
 import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def main():
