@@ -291,7 +291,10 @@ class EcContext:
         parity arrays. Mirrors encode_chunks marshalling
         (ErasureCodeJerasure.cc:121-164)."""
         sizes = {d.nbytes for d in data if d is not None}
-        assert len(sizes) == 1
+        if chunk_bytes is not None:
+            sizes.add(chunk_bytes)
+        assert len(sizes) == 1, "equal-length chunks required (pass " \
+            "chunk_bytes when every chunk is a zeros sentinel)"
         n = sizes.pop()
         if out is None:
             out = [np.zeros(n, dtype=np.uint8) for _ in range(self.m)]

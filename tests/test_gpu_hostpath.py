@@ -112,3 +112,18 @@ def test_legacy_path_still_correct(mods, monkeypatch):
                        text=True, env={**__import__('os').environ,
                                         "ECX_HOSTPIPE": "0"})
     assert r.returncode == 0 and "LEGACY_OK" in r.stdout, r.stderr
+
+
+def test_all_zeros_sentinels_with_explicit_chunk_bytes(mods):
+    """Every chunk a zeros sentinel (None): the wrapper cannot infer the
+    length, so the explicit chunk_bytes parameter carries it (edge found
+    by tools/fuzz_gpu.py); parity of all-zeros data is all zeros
+    (the zeroinout property)."""
+    ceph_amd, _ = mods
+    ctx = ceph_amd.EcContext(4, 2, "reed_sol_van", device=0)
+    try:
+        par = ctx.encode_chunks([None] * 4, chunk_bytes=65536)
+        for p in par:
+            assert p.nbytes == 65536 and not p.any()
+    finally:
+        ctx.close()

@@ -59,7 +59,7 @@ def main():
                 want = oracle.encode_w16(k, m, full)
             else:
                 want = oracle.encode(tech, k, m, full)
-            got = ctx.encode_chunks(data)
+            got = ctx.encode_chunks(data, chunk_bytes=C)
             for j in range(m):
                 assert np.array_equal(got[j], want[j]), ("enc", repro, j)
             ne = int(rng.integers(1, m + 1))
