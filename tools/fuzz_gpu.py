@@ -16,7 +16,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=float, default=240)
-    ap.add_argument("--seed", type=int, default=0xF0)
+    ap.add_argument("--seed", type=lambda x: int(x, 0), default=0xF0)
     args = ap.parse_args()
 
     import ceph_amd
