@@ -152,6 +152,18 @@ int main(int argc, char **argv) {
     CHECK(min2.size() == 4, "minimum == first k available on erasure");
   }
 
+  {
+    // preload: the OSD's osd_erasure_code_plugins startup list
+    // (ErasureCodePlugin.cc preload: csv of names, load+register each)
+    std::stringstream pss;
+    int r = reg.preload("oracle,lrc", dir, &pss);
+    CHECK(r == 0, "preload csv loads and registers each plugin");
+    CHECK(reg.get("oracle") != nullptr && reg.get("lrc") != nullptr,
+          "preloaded plugins visible in the registry");
+    r = reg.preload("no_such_plugin", dir, &pss);
+    CHECK(r != 0, "preload propagates a load failure");
+  }
+
   std::cout << (failures ? "FAILURES: " : "all ok: ") << failures << "\n";
   return failures ? 1 : 0;
 }
