@@ -154,12 +154,18 @@ int main(int argc, char **argv) {
 
   {
     // preload: the OSD's osd_erasure_code_plugins startup list
-    // (ErasureCodePlugin.cc preload: csv of names, load+register each)
+    // (ErasureCodePlugin.cc:208-224 calls load() per csv name). shec and
+    // clay are not yet loaded here => fresh preload succeeds; preloading
+    // an already-registered name goes through load()->init()->add() and
+    // returns -EEXIST, exactly as the reference would (preload is a
+    // startup-time call, before any factory)
     std::stringstream pss;
-    int r = reg.preload("oracle,lrc", dir, &pss);
-    CHECK(r == 0, "preload csv loads and registers each plugin");
-    CHECK(reg.get("oracle") != nullptr && reg.get("lrc") != nullptr,
+    int r = reg.preload("shec,clay", dir, &pss);
+    CHECK(r == 0, "preload csv loads and registers each fresh plugin");
+    CHECK(reg.get("shec") != nullptr && reg.get("clay") != nullptr,
           "preloaded plugins visible in the registry");
+    r = reg.preload("oracle", dir, &pss);
+    CHECK(r == -EEXIST, "re-preloading a registered plugin => -EEXIST");
     r = reg.preload("no_such_plugin", dir, &pss);
     CHECK(r != 0, "preload propagates a load failure");
   }
