@@ -15,8 +15,13 @@ HARNESS = os.path.join(ROOT, "ceph_amd", "harness")
 CORPUS = os.path.join(ROOT, "tests", "golden", "corpus")
 TOOL = os.path.join(HARNESS, "ec_non_regression")
 
-CONFIGS = sorted(d for d in os.listdir(CORPUS)
-                 if d.startswith("plugin=oracle "))
+ALL_CONFIGS = sorted(d for d in os.listdir(CORPUS)
+                     if d.startswith("plugin="))
+CONFIGS = [d for d in ALL_CONFIGS if d.startswith("plugin=oracle ")]
+
+
+def plugin_of(dirname):
+    return dirname.split(" ")[0].split("=", 1)[1]
 
 
 def args_of(dirname):
@@ -40,9 +45,12 @@ def run_check(base, plugin, dirname):
          *args_of(dirname), "--check"], capture_output=True, text=True)
 
 
-@pytest.mark.parametrize("dirname", CONFIGS)
-def test_oracle_matches_committed_corpus(dirname):
-    r = run_check(CORPUS, "oracle", dirname)
+@pytest.mark.parametrize("dirname", ALL_CONFIGS)
+def test_cpu_plugins_match_committed_corpus(dirname):
+    """Every committed corpus directory re-checks with the plugin that
+    wrote it (oracle for the base techniques; lrc/clay compose their CPU
+    sub-plugins)."""
+    r = run_check(CORPUS, plugin_of(dirname), dirname)
     assert r.returncode == 0, (dirname, r.stderr)
 
 
