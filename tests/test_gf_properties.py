@@ -56,3 +56,32 @@ def test_gf8_embeds_in_operations():
         if x == 1:
             break
     assert n == 65535
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.sampled_from(["reed_sol_van", "cauchy", "jerasure_reed_sol_van"]),
+       st.integers(min_value=2, max_value=12),
+       st.integers(min_value=1, max_value=4),
+       st.integers(min_value=1, max_value=8),
+       st.randoms(use_true_random=False))
+def test_oracle_roundtrip_random_shapes(tech, k, m, blocks, rnd):
+    """Encode -> erase up to m chunks -> decode reproduces the stripe,
+    over randomized shapes and erasure patterns (hypothesis-driven CPU
+    companion of the GPU fuzz sweep)."""
+    import numpy as np
+    C = 16 * blocks
+    rng = np.random.default_rng(rnd.randrange(2**32))
+    data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+    par = oracle.encode(tech, k, m, data)
+    n = k + m
+    ne = rnd.randrange(1, m + 1)
+    erased = sorted(rnd.sample(range(n), ne))
+    chunks = [d.copy() for d in data] + [p.copy() for p in par]
+    ref = data + par
+    present = np.ones(n, np.uint8)
+    for e in erased:
+        present[e] = 0
+        chunks[e][:] = 0
+    oracle.decode(tech, k, m, chunks, present)
+    for i in range(n):
+        assert (chunks[i] == ref[i]).all(), (tech, k, m, erased, i)
