@@ -93,3 +93,22 @@ def test_layers_with_object_profiles():
                   "-s", "65536", "-i", "2", "-w", "decode", "-e", "1",
                   "-E", "exhaustive")
     assert r.returncode == 0, r.stderr + r.stdout
+
+
+@pytest.mark.parametrize("mapping,layers", [
+    # mapping length != layer-string length (ErasureCodeLrc parse checks)
+    ("DD__DD__", '[ [ "DDc_", "" ] ]'),
+    # layers value not a JSON array
+    ("DD__", '{"not": "an array"}'),
+    # layer entry not a pair
+    ("DD__", '[ [ "DD__" ] ]'),
+    # unknown symbol in the layer spec
+    ("DD__", '[ [ "DDxq", "" ] ]'),
+])
+def test_invalid_layer_profiles_rejected(mapping, layers):
+    """Malformed mapping/layers profiles must be rejected at init
+    (ErasureCodeLrc.cc layer parsing error paths)."""
+    r = run_bench("-p", "lrc", "-P", "lrc-default-plugin=oracle",
+                  "-P", f"mapping={mapping}", "-P", f"layers={layers}",
+                  "-s", "65536", "-i", "1")
+    assert r.returncode != 0, (mapping, layers, r.stdout)
