@@ -33,8 +33,15 @@
  * parity-delta equivalences), (c) documented structural facts (RS-van row k
  * is all-ones => first parity is the XOR of data, relied on by
  * ErasureCodeIsa.cc:395-456), and (d) committed golden vectors under
- * tests/golden/ generated once by this oracle (self-pin against regression).
- * Before claiming bit-exactness vs a real Ceph install, spot-verify there.
+ * tests/golden/ generated once by this oracle (self-pin against regression),
+ * (e) a closed-form uniqueness cross-check of the Vandermonde construction
+ * (tests/test_gf_kat.py: the systematic form of a fixed code is unique, so
+ * the restatement is fully determined by the published spec), and (f) a
+ * committed non-regression chunk corpus (tests/golden/corpus, the format of
+ * ceph_erasure_code_non_regression.cc) replayed by both the oracle plugin
+ * (CPU) and the GPU plugin every round. Before claiming bit-exactness vs a
+ * real Ceph install, replay a corpus generated there with
+ * ceph_amd/harness/ec_non_regression --check.
  */
 #ifndef EC_REF_H
 #define EC_REF_H
