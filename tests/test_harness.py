@@ -114,3 +114,14 @@ def test_claimed_flags_match_verified_behaviour(plugin, profile, expected):
     r = run_bench("-p", plugin, *profile, "--flags")
     assert r.returncode == 0, r.stderr
     assert r.stdout.strip() == expected
+
+
+@pytest.mark.parametrize("width", [1, 15, 17, 4095])
+def test_tiny_and_odd_stripe_widths(width):
+    """encode() pads arbitrary input widths to the per-technique chunk
+    size (ErasureCode::encode_prepare, ErasureCode.cc:277-312); every
+    width must round-trip through exhaustive single-erasure decode."""
+    r = run_bench("-p", "oracle", "-P", "technique=reed_sol_van",
+                  "-P", "k=3", "-P", "m=2", "-s", str(width), "-i", "1",
+                  "-w", "decode", "-e", "1", "-E", "exhaustive")
+    assert r.returncode == 0, (width, r.stderr)
