@@ -1569,6 +1569,15 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
 int ecx_set_matrix(ecx_ctx* ctx, const uint8_t* coding_rows) {
   if (!ctx || !coding_rows || ctx->is_bitmatrix() || ctx->is_w16())
     return ECX_ERR_INVAL;
+  // Take EVERY slot mutex before mutating ctx->gen: encode paths read the
+  // generator under only their own slot mutex, so rewriting it under
+  // lru_mu alone could expose a torn matrix to a concurrent encode on
+  // another slot. Slot locks are acquired in slot order (the only place
+  // more than one is held), so this cannot deadlock against per-call
+  // single-slot locking.
+  std::vector<std::unique_lock<std::recursive_mutex>> slot_locks;
+  slot_locks.reserve(ctx->slots.size());
+  for (auto& s : ctx->slots) slot_locks.emplace_back(s.mu);
   std::lock_guard<std::mutex> g(ctx->lru_mu);
   std::memcpy(ctx->gen.data() + (size_t)ctx->k * ctx->k, coding_rows,
               (size_t)ctx->m * ctx->k);
@@ -1576,10 +1585,7 @@ int ecx_set_matrix(ecx_ctx* ctx, const uint8_t* coding_rows) {
   ctx->lru_order.clear();
   ctx->bit_lru.clear();
   ctx->lru16.clear();
-  for (auto& s : ctx->slots) {
-    std::lock_guard<std::recursive_mutex> sg(s.mu);
-    s.pparams_kind = 0;  // resident encode tables now stale
-  }
+  for (auto& s : ctx->slots) s.pparams_kind = 0;  // resident tables stale
   return ECX_OK;
 }
 

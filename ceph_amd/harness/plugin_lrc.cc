@@ -292,7 +292,12 @@ class ErasureCodeLrc final : public ErasureCode {
     shard_id_set erasures;
     for (auto &&[shard, b] : out) { (void)b; erasures.insert(shard); }
 
+    // seed with the wanted erasures so that a decode where EVERY layer is
+    // skipped (e.g. a raw caller supplied too few buffers) reports -EIO
+    // instead of silently returning unrecovered chunks
     shard_id_set want_to_read_erasures;
+    for (auto s : want_to_read)
+      if (erasures.contains(s)) want_to_read_erasures.insert(s);
     for (auto layer = layers.rbegin(); layer != layers.rend(); ++layer) {
       shard_id_set layer_erasures;
       for (auto s : layer->chunk_set)
@@ -301,6 +306,19 @@ class ErasureCodeLrc final : public ErasureCode {
               layer->erasure_code->get_coding_chunk_count() ||
           layer_erasures.empty())
         continue;
+      // A raw decode_chunks caller (not ErasureCode::_decode, which
+      // invents buffers for every chunk) may omit buffers for chunks it
+      // considers unavailable; a layer whose survivors are not all backed
+      // by a buffer cannot run — skip it instead of crashing on .at()
+      bool layer_backed = true;
+      for (int c : layer->chunks) {
+        shard_id_t cs((int8_t)c);
+        if (!erasures.contains(cs) && !in.contains(cs) && !out.contains(cs)) {
+          layer_backed = false;
+          break;
+        }
+      }
+      if (!layer_backed) continue;
       shard_id_set layer_want;
       shard_id_map<buffer> layer_in(get_chunk_count());
       shard_id_map<buffer> layer_out(get_chunk_count());

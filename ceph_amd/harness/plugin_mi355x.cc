@@ -169,17 +169,23 @@ class ErasureCodeMi355x final : public ErasureCode {
                     shard_id_map<buffer> &in,
                     shard_id_map<buffer> &out) override {
     (void)want_to_read;  // all erasures in `out` are reconstructed
+    // all in/out buffer lengths must agree (mirror of the encode_chunks
+    // checks; the reference asserts equal blocksize per call,
+    // ErasureCodeJerasure.cc:226-246) — mismatched lengths would reach
+    // ecx_decode_chunks_host, which memcpys `size` bytes per chunk
     size_t size = 0;
     uint8_t *chunks[64] = {};
     uint64_t present = 0;
     std::vector<buffer> temps;
     for (auto &&[shard, b] : in) {
-      size = b.length();
+      if (!size) size = b.length();
+      else if (size != b.length()) return -EINVAL;
       chunks[(int)shard] = b.c_str();
       present |= 1ull << (int)shard;
     }
     for (auto &&[shard, b] : out) {
-      size = b.length();
+      if (!size) size = b.length();
+      else if (size != b.length()) return -EINVAL;
       chunks[(int)shard] = b.c_str();
       present &= ~(1ull << (int)shard);
     }

@@ -21,6 +21,39 @@ using namespace ecx;
 
 namespace {
 
+// Shared by the plugin's _minimum_to_decode and the CPU test probe below:
+// the reference's locality-aware minimum (ErasureCodeShec.cc:130-178 with
+// the construction at :943-962). Takes the FULL want mask — the
+// want&&avail terms of the construction add wanted available chunks.
+int shec_minimum_mask(const std::vector<uint8_t> &coding, int k, int m,
+                      uint64_t want, uint64_t avail, uint64_t *out_mask) {
+  if ((want & avail) == want) {
+    *out_mask = want;
+    return 0;
+  }
+  ShecPlan plan;
+  if (!ecx::shec_decode_plan(coding, k, m, want, avail, plan)) return -EIO;
+  uint64_t mask = 0;
+  for (int id : plan.minimum) mask |= 1ull << id;
+  *out_mask = mask;
+  return 0;
+}
+
+}  // namespace
+
+// CPU-only probe for tests (no GPU context needed): builds the shingled
+// matrix for (k,m,c) and runs the exact minimum computation the plugin's
+// _minimum_to_decode uses. dlopen-able from pytest on a GPU-less box.
+extern "C" int ecx_shec_minimum_probe(int k, int m, int c, int single,
+                                      uint64_t want_mask, uint64_t avail_mask,
+                                      uint64_t *minimum_mask) {
+  std::vector<uint8_t> coding;
+  if (!ecx::shec_matrix(coding, k, m, c, single != 0)) return -EINVAL;
+  return shec_minimum_mask(coding, k, m, want_mask, avail_mask, minimum_mask);
+}
+
+namespace {
+
 class ErasureCodeShec final : public ErasureCode {
   ecx_ctx *ctx_ = nullptr;
   int k_ = 0, m_ = 0, c_ = 0, w_ = 8, device_ = 0, streams_ = 2;
@@ -176,14 +209,15 @@ class ErasureCodeShec final : public ErasureCode {
                          shard_id_set *minimum) override {
     uint64_t want = 0, avail = available.low_mask();
     for (auto &&s : want_to_read) want |= 1ull << (int)s;
-    if ((want & avail) == want) {
-      *minimum = want_to_read;
-      return 0;
-    }
-    ShecPlan plan;
-    if (!ecx::shec_decode_plan(coding_, k_, m_, want & ~avail, avail, plan))
-      return -EIO;
-    for (int id : plan.minimum) minimum->insert(id);
+    // FULL want mask (not want & ~avail): the reference's minimum
+    // construction (ErasureCodeShec.cc:943-962, mirrored in gf.cpp) adds
+    // wanted-but-available chunks via its want[i]&&avails[i] terms; the
+    // plan search itself only uses want&&!avail terms, so this is safe.
+    uint64_t min_mask = 0;
+    int r = shec_minimum_mask(coding_, k_, m_, want, avail, &min_mask);
+    if (r) return r;
+    for (int i = 0; i < k_ + m_; i++)
+      if (min_mask & (1ull << i)) minimum->insert(i);
     return 0;
   }
 

@@ -153,6 +153,76 @@ int main(int argc, char **argv) {
   }
 
   {
+    // LRC decode_chunks from a RAW caller (not ErasureCode::_decode, which
+    // invents buffers for all k+m chunks): chunks in neither map must not
+    // crash the plugin — layers that lack a backing buffer are skipped and
+    // the call returns 0 (recovered) or -EIO, never throws
+    ErasureCodeProfile p{{"k", "4"}, {"m", "2"}, {"l", "3"},
+                         {"lrc-default-plugin", "oracle"}};
+    ErasureCodeInterfaceRef lrc;
+    int r = reg.factory("lrc", dir, p, &lrc, &ss);
+    CHECK(r == 0, "lrc factory (oracle sub-plugin)");
+    if (r == 0) {
+      const unsigned n = lrc->get_chunk_count();
+      const unsigned K = lrc->get_data_chunk_count();
+      const unsigned C = lrc->get_chunk_size(4096 * 4);
+      // raw positions: chunk_mapping[logical shard] = raw chunk index;
+      // data occupies the mapping's 'D' positions, not 0..K-1
+      auto &cm = lrc->get_chunk_mapping();
+      std::vector<int> data_pos;
+      std::vector<bool> is_data(n, false);
+      for (unsigned i = 0; i < K; i++) {
+        data_pos.push_back((int)cm[i]);
+        is_data[(int)cm[i]] = true;
+      }
+      std::mt19937 rng(0xECEC);
+      shard_id_map<buffer> enc_in(n), enc_out(n);
+      for (int dp : data_pos) {
+        buffer b = buffer::create_aligned(C);
+        for (unsigned x = 0; x < C; x++) b.c_str()[x] = (uint8_t)rng();
+        enc_in[dp] = b;
+      }
+      for (unsigned i = 0; i < n; i++)
+        if (!is_data[i]) enc_out[i] = buffer::create_aligned(C);
+      CHECK(lrc->encode_chunks(enc_in, enc_out) == 0, "lrc encode for raw test");
+      const int d0 = data_pos[0];
+      // erase data chunk d0; supply buffers for only two other data
+      // chunks — every layer containing d0 also contains unbacked chunks
+      shard_id_set want;
+      want.insert(d0);
+      shard_id_map<buffer> in(n), out(n);
+      in[data_pos[1]] = enc_in.at(data_pos[1]);
+      in[data_pos[2]] = enc_in.at(data_pos[2]);
+      out[d0] = buffer::create_aligned(C);
+      bool threw = false;
+      int dr = 0;
+      try {
+        dr = lrc->decode_chunks(want, in, out);
+      } catch (...) {
+        threw = true;
+      }
+      CHECK(!threw, "lrc raw decode_chunks with missing buffers: no throw");
+      // either the plugin reports failure, or (if some backed layer could
+      // recover) the result must be byte-exact — never a crash or garbage
+      CHECK(threw || dr != 0 ||
+                !std::memcmp(out.at(d0).c_str(), enc_in.at(d0).c_str(), C),
+            "lrc raw decode: error or exact recovery, nothing in between");
+      // with every other chunk backed, recovery succeeds and is exact
+      shard_id_map<buffer> in2(n), out2(n);
+      for (unsigned i = 0; i < n; i++) {
+        if ((int)i == d0) continue;
+        if (enc_in.contains(i)) in2[i] = enc_in.at(i);
+        else in2[i] = enc_out.at(i);
+      }
+      out2[d0] = buffer::create_aligned(C);
+      int dr2 = lrc->decode_chunks(want, in2, out2);
+      CHECK(dr2 == 0 && !std::memcmp(out2.at(d0).c_str(),
+                                     enc_in.at(d0).c_str(), C),
+            "lrc raw decode with full buffers recovers the erased chunk");
+    }
+  }
+
+  {
     // preload: the OSD's osd_erasure_code_plugins startup list
     // (ErasureCodePlugin.cc:208-224 calls load() per csv name). shec and
     // clay are not yet loaded here => fresh preload succeeds; preloading

@@ -148,3 +148,65 @@ def test_shec_plugin_encode_runs():
     r = run_bench("-p", "shec", "-P", "k=6", "-P", "m=4", "-P", "c=2",
                   "-s", str(1 << 20), "-i", "3")
     assert r.returncode == 0, r.stderr
+
+
+# ---- minimum_to_decode (CPU, via the ecx_shec_minimum_probe export) ----
+# The probe in plugin_shec.cc runs the exact minimum computation the
+# plugin's _minimum_to_decode uses (ErasureCodeShec.cc:130-178 + :943-962).
+
+def _shec_minimum(k, m, c, want_ids, avail_ids, single=False):
+    import ctypes
+    lib = ctypes.CDLL(os.path.join(HARNESS, "libec_shec.so"))
+    fn = lib.ecx_shec_minimum_probe
+    fn.restype = ctypes.c_int
+    fn.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                   ctypes.c_uint64, ctypes.c_uint64,
+                   ctypes.POINTER(ctypes.c_uint64)]
+    want = sum(1 << i for i in want_ids)
+    avail = sum(1 << i for i in avail_ids)
+    out = ctypes.c_uint64(0)
+    rc = fn(k, m, c, int(single), want, avail, ctypes.byref(out))
+    return rc, {i for i in range(64) if out.value >> i & 1}
+
+
+def test_minimum_all_wanted_available():
+    rc, mn = _shec_minimum(4, 3, 2, {0, 2}, {0, 1, 2, 3, 4})
+    assert rc == 0 and mn == {0, 2}
+
+
+def test_minimum_includes_wanted_available_data():
+    """Regression for the want&~avail masking bug: when chunk 1 is erased
+    and chunks {0,1} are wanted, the minimum must still include the
+    wanted AVAILABLE chunk 0 (ErasureCodeShec.cc:957-959 adds
+    want[i]&&avails[i] chunks)."""
+    k, m, c = 4, 3, 2
+    avail = set(range(k + m)) - {1}
+    rc, mn = _shec_minimum(k, m, c, {0, 1}, avail)
+    assert rc == 0
+    assert 0 in mn, mn
+    # minimum is a read set: every member must be available
+    assert mn <= avail, mn
+    # and it must contain enough chunks to actually recover chunk 1
+    assert len(mn - {0}) >= 1
+
+
+@pytest.mark.parametrize("erased", range(7))
+def test_minimum_single_erasure_subset_of_available(erased):
+    k, m, c = 4, 3, 2
+    avail = set(range(k + m)) - {erased}
+    want = {erased} | ({0} if erased != 0 else {2})
+    rc, mn = _shec_minimum(k, m, c, want, avail)
+    assert rc == 0
+    assert mn <= avail, (erased, mn)
+    assert (want & avail) <= mn, (erased, mn)
+
+
+def test_minimum_wanted_available_parity_kept():
+    """A wanted available parity chunk is added to minimum when it covers
+    a data chunk outside want (ErasureCodeShec.cc:943-953 branch)."""
+    k, m, c = 4, 3, 2
+    avail = set(range(k + m)) - {0}
+    rc, mn = _shec_minimum(k, m, c, {0, k}, avail)  # want data 0 + parity k
+    assert rc == 0
+    assert k in mn, mn
+    assert mn <= avail, mn
