@@ -18,12 +18,14 @@ T_CAUCHY_ISA = 1
 T_RS_VAN_JERASURE = 2
 T_CAUCHY_ORIG_JERASURE = 3
 T_RS_VAN_JERASURE_W16 = 4
+T_CAUCHY_GOOD_JERASURE = 5
 TECHNIQUES = {
     "reed_sol_van": T_RS_VAN_ISA,
     "cauchy": T_CAUCHY_ISA,
     "jerasure_reed_sol_van": T_RS_VAN_JERASURE,
     "cauchy_orig": T_CAUCHY_ORIG_JERASURE,
     "jerasure_reed_sol_van_w16": T_RS_VAN_JERASURE_W16,
+    "cauchy_good": T_CAUCHY_GOOD_JERASURE,
 }
 
 _ERR = {

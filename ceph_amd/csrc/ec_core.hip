@@ -446,7 +446,7 @@ struct EcBitParams {
   // blob continues with uint16 ops[row_off[n_rows]]: values j*w+c
 };
 
-template <bool NT>
+template <bool NT, bool ACCUM = false>
 __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
@@ -488,7 +488,12 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
     const int r = t >> vq_shift;
     const int v = t - (r << vq_shift);
-    v4u acc = {0, 0, 0, 0};
+    v4u* dst = reinterpret_cast<v4u*>(
+        obase + (long)bp->out_ids[r / w] * chunk_bytes + sw * (long)w * pkt +
+        (long)(r % w) * pkt + (long)win * q + (long)v * 16);
+    // ACCUM = parity-delta apply (schedule_apply_delta semantics,
+    // ErasureCodeJerasure.cc:348-377): XOR into the existing parity
+    v4u acc = ACCUM ? *dst : v4u{0, 0, 0, 0};
     const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
     for (int o = b0; o < b1; o++) {
       const int jc = s_ops[o];
@@ -496,9 +501,6 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
           s_data + (size_t)jc * q + (size_t)v * 16);
       acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
     }
-    v4u* dst = reinterpret_cast<v4u*>(
-        obase + (long)bp->out_ids[r / w] * chunk_bytes + sw * (long)w * pkt +
-        (long)(r % w) * pkt + (long)win * q + (long)v * 16);
     if (NT)
       __builtin_nontemporal_store(acc, dst);
     else
@@ -616,7 +618,10 @@ struct ecx_ctx {
   // stream's mutex
   std::atomic<unsigned> rr{0};
 
-  bool is_bitmatrix() const { return technique == ECX_T_CAUCHY_ORIG_JERASURE; }
+  bool is_bitmatrix() const {
+    return technique == ECX_T_CAUCHY_ORIG_JERASURE ||
+           technique == ECX_T_CAUCHY_GOOD_JERASURE;
+  }
   bool is_w16() const { return technique == ECX_T_RS_VAN_JERASURE_W16; }
 };
 
@@ -649,9 +654,14 @@ int ecx_create2(int k, int m, int technique, int w, int packetsize,
   if (!out || k < 2 || m < 1 || k > ECX_MAX_K || m > ECX_MAX_K ||
       n_streams < 1 || n_streams > 64 || w != (want16 ? 16 : 8))
     return ECX_ERR_INVAL;
-  if (technique == ECX_T_CAUCHY_ORIG_JERASURE &&
+  if ((technique == ECX_T_CAUCHY_ORIG_JERASURE ||
+       technique == ECX_T_CAUCHY_GOOD_JERASURE) &&
       (packetsize < 16 || packetsize % 16 || k > 16))
     return ECX_ERR_INVAL;
+  // cauchy_good m==2: jerasure would use its precomputed cbest tables
+  // (cauchy.c), which cannot be faithfully restated here — refuse rather
+  // than silently diverge from the reference's parity bytes.
+  if (technique == ECX_T_CAUCHY_GOOD_JERASURE && m == 2) return ECX_ERR_INVAL;
   if (ecx_device_count() <= device) return ECX_ERR_NO_GPU;
 
   auto ctx = new ecx_ctx();
@@ -1163,7 +1173,8 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
                          uint8_t* d_obuf, const int* src_ids, int n_src,
                          const int* out_ids, int n_out,
                          const uint8_t* bit_rows, long n_stripes,
-                         size_t chunk_bytes, bool time_it) {
+                         size_t chunk_bytes, bool time_it,
+                         bool accum = false) {
   const int w = ctx->w, pkt = ctx->pkt;
   if (n_src < 1 || n_src > 16 || n_out < 1 || n_out > ECX_MAX_OUT)
     return ECX_ERR_INVAL;
@@ -1235,14 +1246,13 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   dim3 grid((unsigned)(sw_per_chunk * windows_per_sw), (unsigned)n_stripes);
   size_t lds = (size_t)n_src * w * q + ((ops.size() * 2 + 15) & ~15ull);
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
-  if (env_nt)
-    hipLaunchKernelGGL((ec_bitmatrix_kernel<true>), grid, dim3(256), lds,
-                       s.stream, d_buf, d_obuf, s.d_jobs, (long)chunk_bytes,
-                       ctx->k + ctx->m, windows_per_sw);
-  else
-    hipLaunchKernelGGL((ec_bitmatrix_kernel<false>), grid, dim3(256), lds,
-                       s.stream, d_buf, d_obuf, s.d_jobs, (long)chunk_bytes,
-                       ctx->k + ctx->m, windows_per_sw);
+  auto kfn = env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
+                             : ec_bitmatrix_kernel<true, false>)
+                    : (accum ? ec_bitmatrix_kernel<false, true>
+                             : ec_bitmatrix_kernel<false, false>);
+  hipLaunchKernelGGL(kfn, grid, dim3(256), lds, s.stream, d_buf, d_obuf,
+                     s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
+                     windows_per_sw);
   HIP_TRY(hipGetLastError());
   if (time_it) {
     HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
@@ -1520,7 +1530,24 @@ int ecx_apply_delta_dev(ecx_ctx* ctx, const void* d_delta, int data_shard,
                         (uint8_t*)d_parity, sids, 1, oids, 1, cf, 1, bytes,
                         true, false);
   }
-  if (ctx->is_bitmatrix()) return ECX_ERR_INVAL;  // schedule deltas: later
+  if (ctx->is_bitmatrix()) {
+    // schedule-delta apply (schedule_apply_delta filtered to one
+    // (datashard, codingshard) pair, ErasureCodeJerasure.cc:348-377):
+    // the pair's w x w bitmatrix block applied per superword, XORed into
+    // the existing parity. Extract block (i, j) as w bit rows over one
+    // source chunk.
+    const int w = ctx->w, W = ctx->k * w;
+    const int i = coding_shard - ctx->k, j = data_shard;
+    std::vector<uint8_t> rows((size_t)w * w);
+    for (int r = 0; r < w; r++)
+      for (int c = 0; c < w; c++)
+        rows[(size_t)r * w + c] =
+            ctx->bitmat[(size_t)(i * w + r) * W + j * w + c];
+    int sids[1] = {0}, oids[1] = {0};
+    return run_bitmatrix(ctx, slot, (const uint8_t*)d_delta,
+                         (uint8_t*)d_parity, sids, 1, oids, 1, rows.data(),
+                         1, bytes, false, /*accum=*/true);
+  }
   uint8_t c = ctx->gen[(size_t)coding_shard * ctx->k + data_shard];
   int src_ids[1] = {0};
   int out_ids[1] = {0};
@@ -1636,6 +1663,23 @@ int ecx_matmul_chunks_host(ecx_ctx* ctx, const uint8_t* const* srcs,
   }
   HIP_TRY(hipStreamSynchronize(s.stream));
   return ECX_OK;
+}
+
+int ecx_gen_matrix_probe(int technique, int k, int m, uint8_t* out) {
+  // CPU-only matrix readback (no GPU context): lets tests pin the core's
+  // generator constructions against the oracle's on a GPU-less box.
+  if (!out || k < 1 || m < 1 || k + m > 64) return ECX_ERR_INVAL;
+  if (technique == ECX_T_CAUCHY_GOOD_JERASURE && m == 2)
+    return ECX_ERR_INVAL;  // cbest tables unsourceable
+  std::vector<uint8_t> a;
+  if (!ecx::gen_matrix(technique, a, k, m)) return ECX_ERR_INVAL;
+  std::memcpy(out, a.data(), a.size());
+  return ECX_OK;
+}
+
+int ecx_cauchy_n_ones_probe(int e) {
+  if (e < 0 || e > 255) return ECX_ERR_INVAL;
+  return ecx::cauchy_n_ones((uint8_t)e);
 }
 
 int ecx_shec_matrix(int k, int m, int c, int single, uint8_t* out) {

@@ -111,8 +111,83 @@ bool gen_matrix(int technique, std::vector<uint8_t> &a, int k, int m) {
     case 1: return gen_matrix_cauchy_isa(a, k, m);
     case 2: return gen_matrix_rs_van_jerasure(a, k, m);
     case 3: return gen_matrix_cauchy_orig(a, k, m);
+    case 5: return gen_matrix_cauchy_good(a, k, m);
     default: return false;
   }
+}
+
+// jerasure cauchy.c cauchy_n_ones semantics, w=8: the number of ones in
+// the 8x8 companion-basis bitmatrix of e. Column c of that block is the
+// bit pattern of e*2^c (matrix_to_bitmatrix above), so the count is
+// sum_{c=0..7} popcount(e * 2^c). (jerasure computes the same value with
+// an incremental shift-and-correct recurrence; this is the direct form.)
+int cauchy_n_ones(uint8_t e) {
+  const GF8 &f = gf8();
+  int no = 0;
+  uint8_t v = e;
+  for (int c = 0; c < 8; c++) {
+    no += __builtin_popcount(v);
+    v = f.mul(v, 2);
+  }
+  return no;
+}
+
+// jerasure cauchy.c cauchy_improve_coding_matrix (the behaviour the
+// in-tree plugin documents for technique=cauchy_good,
+// ErasureCodeJerasure.cc:537-555), operating on the m x k coding rows:
+//  1) divide each COLUMN j by its row-0 element, making row 0 all ones
+//     (an all-ones row bitmatrix is pure XOR — zero multiply cost);
+//  2) for each row i >= 1, scan elements: dividing the row by element j
+//     yields total bitmatrix ones tno = sum_col n_ones(row[col]/row[j]);
+//     take the first j strictly improving the current count, then divide
+//     the whole row by that element.
+// Row/column scaling by nonzero constants preserves the MDS property of
+// the Cauchy matrix (every square submatrix determinant scales by a
+// nonzero factor).
+void improve_cauchy_matrix(uint8_t *coding, int k, int m) {
+  const GF8 &f = gf8();
+  for (int j = 0; j < k; j++) {
+    uint8_t e = coding[j];
+    if (e != 1) {
+      uint8_t tmp = f.div(1, e);
+      for (int i = 1; i < m; i++)
+        coding[(size_t)i * k + j] = f.mul(coding[(size_t)i * k + j], tmp);
+      coding[j] = 1;
+    }
+  }
+  for (int i = 1; i < m; i++) {
+    uint8_t *row = coding + (size_t)i * k;
+    int bno = 0;
+    for (int j = 0; j < k; j++) bno += cauchy_n_ones(row[j]);
+    int bno_index = -1;
+    for (int j = 0; j < k; j++) {
+      if (row[j] == 1) continue;
+      uint8_t tmp = f.div(1, row[j]);
+      int tno = 0;
+      for (int col = 0; col < k; col++)
+        tno += cauchy_n_ones(f.mul(row[col], tmp));
+      if (tno < bno) {
+        bno = tno;
+        bno_index = j;
+      }
+    }
+    if (bno_index != -1) {
+      uint8_t tmp = f.div(1, row[bno_index]);
+      for (int j = 0; j < k; j++) row[j] = f.mul(row[j], tmp);
+    }
+  }
+}
+
+// jerasure cauchy.c cauchy_good_general_coding_matrix, general branch:
+// cauchy_original + improve. NOTE: for m == 2 (RAID-6) jerasure instead
+// reads precomputed "cbest" element tables whose values cannot be
+// faithfully restated from material available in this container; callers
+// that need jerasure-bit-exact m=2 cauchy_good must gate on that
+// (plugin_mi355x rejects m==2 by default).
+bool gen_matrix_cauchy_good(std::vector<uint8_t> &a, int k, int m) {
+  if (!gen_matrix_cauchy_orig(a, k, m)) return false;
+  improve_cauchy_matrix(a.data() + (size_t)k * k, k, m);
+  return true;
 }
 
 // jerasure cauchy.c cauchy_original_coding_matrix: coding (i,j) =

@@ -52,6 +52,14 @@ bool gf_invert(const uint8_t *in, uint8_t *out, int k);
 // jerasure cauchy.c / jerasure.c semantics as called from
 // ErasureCodeJerasure.cc:499-514,568-574.
 bool gen_matrix_cauchy_orig(std::vector<uint8_t> &a, int k, int m);
+// jerasure cauchy.c cauchy_n_ones (w=8): ones in the companion bitmatrix
+// of e = sum_c popcount(e * 2^c).
+int cauchy_n_ones(uint8_t e);
+// jerasure cauchy.c cauchy_improve_coding_matrix over m x k coding rows.
+void improve_cauchy_matrix(uint8_t *coding, int k, int m);
+// cauchy_good general branch: cauchy_original + improve (m == 2 would use
+// jerasure's unsourceable cbest tables — callers gate on that).
+bool gen_matrix_cauchy_good(std::vector<uint8_t> &a, int k, int m);
 // bitmat: (m*w) x (k*w) bits, one byte per bit, row-major
 void matrix_to_bitmatrix(const uint8_t *coding_rows, int k, int m, int w,
                          std::vector<uint8_t> &bitmat);

@@ -23,6 +23,7 @@ int technique_id(const std::string &t) {
   if (t == "cauchy") return ECX_T_CAUCHY_ISA;
   if (t == "jerasure_reed_sol_van") return ECX_T_RS_VAN_JERASURE;
   if (t == "cauchy_orig") return ECX_T_CAUCHY_ORIG_JERASURE;
+  if (t == "cauchy_good") return ECX_T_CAUCHY_GOOD_JERASURE;
   // jerasure reed_sol_r6_op (RAID6, m==2): its reed_sol_r6_coding_matrix
   // is row0 = all ones, row1 = [2^j] — byte-identical to the isa RS-van
   // construction at m=2 (gf_gen_rs_matrix rows k, k+1)
@@ -35,7 +36,8 @@ class ErasureCodeMi355x final : public ErasureCode {
   int k_ = 0, m_ = 0, w_ = 8, device_ = 0, streams_ = 2, packetsize_ = 2048;
   std::string technique_;
   bool is_bitmatrix() const {
-    return technique_id(technique_) == ECX_T_CAUCHY_ORIG_JERASURE;
+    int t = technique_id(technique_);
+    return t == ECX_T_CAUCHY_ORIG_JERASURE || t == ECX_T_CAUCHY_GOOD_JERASURE;
   }
 
  public:
@@ -56,14 +58,15 @@ class ErasureCodeMi355x final : public ErasureCode {
     // (ErasureCodeJerasure.h:52-63) claim OPTIMIZED only for reed_sol_van
     // and CRC for everything except reed_sol_van and cauchy_orig. The CRC
     // flag is a capability declaration the OSD's scrub consumes — there
-    // is no plugin-side CRC method to implement. Parity-delta is
-    // matrix-technique-only here (cauchy_orig delta schedules are a later
-    // round — claim only what works, the conformance rule).
+    // is no plugin-side CRC method to implement. Parity-delta is claimed
+    // for every technique, as upstream does (matrix deltas via
+    // matrix_apply_delta, bitmatrix deltas via schedule_apply_delta —
+    // both implemented here; delta == re-encode is conformance-tested).
     plugin_flags f = FLAG_EC_PLUGIN_PARTIAL_READ_OPTIMIZATION |
                      FLAG_EC_PLUGIN_PARTIAL_WRITE_OPTIMIZATION |
                      FLAG_EC_PLUGIN_ZERO_INPUT_ZERO_OUTPUT_OPTIMIZATION |
+                     FLAG_EC_PLUGIN_PARITY_DELTA_OPTIMIZATION |
                      FLAG_EC_PLUGIN_DIRECT_READS;
-    if (!is_bitmatrix()) f |= FLAG_EC_PLUGIN_PARITY_DELTA_OPTIMIZATION;
     if (technique_ == "reed_sol_van" || technique_ == "cauchy") {
       // isa-matrix techniques
       f |= FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED;
@@ -72,8 +75,10 @@ class ErasureCodeMi355x final : public ErasureCode {
         f |= FLAG_EC_PLUGIN_CRC_ENCODE_DECODE_SUPPORT;
     } else if (technique_ == "jerasure_reed_sol_van") {
       f |= FLAG_EC_PLUGIN_OPTIMIZED_SUPPORTED;  // jerasure reed_sol_van
-    } else if (!is_bitmatrix()) {
-      f |= FLAG_EC_PLUGIN_CRC_ENCODE_DECODE_SUPPORT;  // e.g. reed_sol_r6_op
+    } else if (technique_ != "cauchy_orig") {
+      // jerasure family: CRC for everything but reed_sol_van/cauchy_orig
+      // (ErasureCodeJerasure.h:57-62) — cauchy_good and reed_sol_r6_op
+      f |= FLAG_EC_PLUGIN_CRC_ENCODE_DECODE_SUPPORT;
     }
     return f;
   }
@@ -100,6 +105,15 @@ class ErasureCodeMi355x final : public ErasureCode {
     }
     if (technique_id(technique_) < 0) {
       if (ss) *ss << "mi355x: unknown technique " << technique_ << "\n";
+      err = -EINVAL;
+    }
+    if (technique_ == "cauchy_good" && m_ == 2) {
+      // jerasure's m==2 cauchy_good reads precomputed cbest tables that
+      // cannot be faithfully restated here; refuse rather than silently
+      // diverge from the reference's parity bytes
+      if (ss)
+        *ss << "cauchy_good: m=2 uses jerasure's cbest tables "
+               "(unsourceable here) — use cauchy_orig or m!=2\n";
       err = -EINVAL;
     }
     if (technique_ == "reed_sol_r6_op" && m_ != 2) {
@@ -211,10 +225,9 @@ class ErasureCodeMi355x final : public ErasureCode {
 
   void apply_delta(const shard_id_map<buffer> &in,
                    shard_id_map<buffer> &out) override {
-    if (is_bitmatrix())
-      throw std::runtime_error(
-          "apply_delta not supported for cauchy_orig (flag not claimed)");
-    // loop structure mirrors isa apply_delta (ErasureCodeIsa.cc:333-366)
+    // loop structure mirrors isa apply_delta (ErasureCodeIsa.cc:333-366);
+    // bitmatrix techniques route to the schedule-delta kernel inside
+    // ecx_apply_delta_host (schedule_apply_delta semantics)
     for (auto &&[datashard, databuf] : in) {
       if ((int)datashard >= k_) continue;
       for (auto &&[codingshard, codingbuf] : out) {

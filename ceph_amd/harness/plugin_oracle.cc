@@ -24,6 +24,7 @@ int technique_id(const std::string &t) {
   if (t == "cauchy") return ECREF_T_CAUCHY_ISA;
   if (t == "jerasure_reed_sol_van") return ECREF_T_RS_VAN_JERASURE;
   if (t == "cauchy_orig") return 3;  // bitmatrix/packet layout
+  if (t == "cauchy_good") return 5;  // bitmatrix (orig + improve pass)
   if (t == "reed_sol_r6_op") return ECREF_T_RS_VAN_ISA;  // RAID6 == isa m=2
   return -1;
 }
@@ -34,7 +35,9 @@ class ErasureCodeOracle final : public ErasureCode {
   std::string technique_;
   std::vector<uint8_t> gen_;
   std::vector<uint8_t> bitmat_;
-  bool is_bitmatrix() const { return technique_ == "cauchy_orig"; }
+  bool is_bitmatrix() const {
+    return technique_ == "cauchy_orig" || technique_ == "cauchy_good";
+  }
 
  public:
   explicit ErasureCodeOracle(std::string t) : technique_(std::move(t)) {}
@@ -74,6 +77,11 @@ class ErasureCodeOracle final : public ErasureCode {
       if (ss) *ss << "oracle: unknown technique " << technique_ << "\n";
       err = -EINVAL;
     }
+    if (technique_ == "cauchy_good" && m_ == 2) {
+      if (ss) *ss << "oracle: cauchy_good m=2 needs jerasure's cbest "
+                     "tables (unsourceable) — refused\n";
+      return -EINVAL;
+    }
     if (technique_ == "reed_sol_r6_op" && m_ != 2) {
       if (ss) *ss << "reed_sol_r6_op: m must be 2\n";
       err = -EINVAL;
@@ -85,7 +93,10 @@ class ErasureCodeOracle final : public ErasureCode {
     gen_.resize((size_t)(k_ + m_) * k_);
     if (is_bitmatrix()) {
       std::vector<uint8_t> coding((size_t)m_ * k_);
-      if (ecref_matrix_cauchy_orig_jerasure(coding.data(), k_, m_) != 0)
+      if ((technique_ == "cauchy_good"
+               ? ecref_matrix_cauchy_good_jerasure(coding.data(), k_, m_)
+               : ecref_matrix_cauchy_orig_jerasure(coding.data(), k_, m_))
+          != 0)
         return -EINVAL;
       for (int i = 0; i < k_; i++) {
         std::memset(&gen_[(size_t)i * k_], 0, k_);

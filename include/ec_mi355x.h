@@ -69,6 +69,13 @@ enum ecx_technique {
    * ecx_create2(..., w=16, ...) with this id. Compute-bound compatibility
    * technique; chunk sizes are multiples of k*w*4 / k = 64 B. */
   ECX_T_RS_VAN_JERASURE_W16 = 4,
+  /* jerasure cauchy_good: cauchy_original + the n_ones-minimising improve
+   * pass (jerasure cauchy.c cauchy_good_general_coding_matrix general
+   * branch; ErasureCodeJerasure.cc:537-555). Same bitmatrix/packet layout
+   * and chunk-size rule as ECX_T_CAUCHY_ORIG_JERASURE. m == 2 is refused
+   * at this layer: jerasure's m==2 path reads precomputed cbest tables
+   * that cannot be faithfully restated here. */
+  ECX_T_CAUCHY_GOOD_JERASURE = 5,
 };
 
 enum ecx_err {
@@ -212,6 +219,12 @@ ECX_API int ecx_matmul_chunks_host(ecx_ctx *ctx,
  * in-tree shec_reedsolomon_coding_matrix (ErasureCodeShec.cc:700-768);
  * single != 0 selects the SINGLE technique. */
 ECX_API int ecx_shec_matrix(int k, int m, int c, int single, uint8_t *out);
+
+/* CPU-only probes (no GPU context; usable on a GPU-less box for tests):
+ * generator readback for any technique id — fills (k+m)*k bytes, identity
+ * top — and jerasure cauchy.c's cauchy_n_ones(e, w=8). */
+ECX_API int ecx_gen_matrix_probe(int technique, int k, int m, uint8_t *out);
+ECX_API int ecx_cauchy_n_ones_probe(int e);
 
 /* Generic device-batch GF(2^8) matmul over the standard batch layout:
  * out chunk ids = XOR_i rows[j*n_src+i] * src chunk ids, per stripe. The

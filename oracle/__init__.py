@@ -17,12 +17,15 @@ T_RS_VAN_ISA = 0
 T_CAUCHY_ISA = 1
 T_RS_VAN_JERASURE = 2
 T_CAUCHY_ORIG_JERASURE = 3  # bitmatrix/packet layout: use bitmatrix_* fns
+T_CAUCHY_GOOD_JERASURE = 5  # bitmatrix/packet layout (orig + improve pass)
 TECHNIQUES = {
     "reed_sol_van": T_RS_VAN_ISA,
     "cauchy": T_CAUCHY_ISA,
     "jerasure_reed_sol_van": T_RS_VAN_JERASURE,
     "cauchy_orig": T_CAUCHY_ORIG_JERASURE,
+    "cauchy_good": T_CAUCHY_GOOD_JERASURE,
 }
+BITMATRIX_TECHNIQUES = (T_CAUCHY_ORIG_JERASURE, T_CAUCHY_GOOD_JERASURE)
 
 
 def _build_if_needed():
@@ -62,6 +65,8 @@ _ref.ecref_chunk_size_jerasure.restype = ctypes.c_uint
 _ref.ecref_chunk_size_jerasure.argtypes = [ctypes.c_int, ctypes.c_int,
                                            ctypes.c_uint]
 _ref.ecref_matrix_cauchy_orig_jerasure.restype = ctypes.c_int
+_ref.ecref_matrix_cauchy_good_jerasure.restype = ctypes.c_int
+_ref.ecref_cauchy_n_ones.restype = ctypes.c_int
 _ref.ecref_bitmatrix_encode.restype = ctypes.c_int
 _ref.ecref_bitmatrix_decode.restype = ctypes.c_int
 _ref.ecref_matrix_rs_vandermonde_jerasure_w16.restype = ctypes.c_int
@@ -90,10 +95,11 @@ def gf_exp_table():
 def matrix(technique, k, m):
     """Full (k+m) x k generator (identity top), as uint8 ndarray."""
     t = TECHNIQUES[technique] if isinstance(technique, str) else technique
-    if t == T_CAUCHY_ORIG_JERASURE:
+    if t in BITMATRIX_TECHNIQUES:
         a = np.zeros((k + m, k), dtype=np.uint8)
         a[:k] = np.eye(k, dtype=np.uint8)
-        a[k:] = cauchy_orig_matrix(k, m)
+        a[k:] = (cauchy_orig_matrix(k, m) if t == T_CAUCHY_ORIG_JERASURE
+                 else cauchy_good_matrix(k, m))
         return a
     a = np.zeros((k + m, k), dtype=np.uint8)
     r = _ref.ecref_matrix(t, a.ctypes.data_as(ctypes.c_void_p), k, m)
@@ -113,8 +119,8 @@ def encode(technique, k, m, data, chunk_bytes=None):
     """data: list of k uint8 arrays (or None for zeros). Returns list of m
     parity arrays. Scalar oracle (ecref_encode)."""
     t = TECHNIQUES[technique] if isinstance(technique, str) else technique
-    if t == T_CAUCHY_ORIG_JERASURE:
-        raise ValueError("cauchy_orig uses the bitmatrix/packet layout: "
+    if t in BITMATRIX_TECHNIQUES:
+        raise ValueError("bitmatrix techniques use the packet layout: "
                          "call bitmatrix_encode/bitmatrix_decode")
     lens = {d.nbytes for d in data if d is not None}
     assert len(lens) == 1 or (not lens and chunk_bytes)
@@ -167,6 +173,30 @@ def cauchy_orig_matrix(k, m):
     return a
 
 
+def cauchy_good_matrix(k, m):
+    """jerasure cauchy_good coding matrix (m x k): cauchy_orig + the
+    n_ones-minimising improve pass (cauchy.c general branch; m==2 would
+    need jerasure's unsourceable cbest tables and raises)."""
+    a = np.zeros((m, k), dtype=np.uint8)
+    r = _ref.ecref_matrix_cauchy_good_jerasure(
+        a.ctypes.data_as(ctypes.c_void_p), k, m)
+    if r != 0:
+        raise ValueError(f"cauchy_good matrix failed: {r}")
+    return a
+
+
+def cauchy_n_ones(e):
+    """Ones in the 8x8 companion bitmatrix of e (cauchy.c cauchy_n_ones)."""
+    return int(_ref.ecref_cauchy_n_ones(ctypes.c_uint8(e)))
+
+
+def _bit_coding_matrix(technique, k, m):
+    t = TECHNIQUES[technique] if isinstance(technique, str) else technique
+    if t == T_CAUCHY_GOOD_JERASURE:
+        return cauchy_good_matrix(k, m)
+    return cauchy_orig_matrix(k, m)
+
+
 def bitmatrix(coding, w=8):
     m, k = coding.shape
     bm = np.zeros((m * w, k * w), dtype=np.uint8)
@@ -176,13 +206,15 @@ def bitmatrix(coding, w=8):
     return bm
 
 
-def bitmatrix_encode(k, m, data, packetsize, w=8):
-    """jerasure cauchy_orig (bitmatrix/packet layout) encode; data entries
-    may be None for zeros chunks."""
+def bitmatrix_encode(k, m, data, packetsize, w=8,
+                     technique=T_CAUCHY_ORIG_JERASURE):
+    """jerasure bitmatrix/packet-layout encode (cauchy_orig by default,
+    technique="cauchy_good" for the improved matrix); data entries may be
+    None for zeros chunks."""
     lens = {d.nbytes for d in data if d is not None}
     assert len(lens) == 1
     size = lens.pop()
-    bm = bitmatrix(cauchy_orig_matrix(k, m), w)
+    bm = bitmatrix(_bit_coding_matrix(technique, k, m), w)
     parity = [np.zeros(size, dtype=np.uint8) for _ in range(m)]
     r = _ref.ecref_bitmatrix_encode(
         k, m, w, bm.ctypes.data_as(ctypes.c_void_p), _ptr_array(data),
@@ -192,8 +224,9 @@ def bitmatrix_encode(k, m, data, packetsize, w=8):
     return parity
 
 
-def bitmatrix_decode(k, m, chunks, present, packetsize, w=8):
-    bm = bitmatrix(cauchy_orig_matrix(k, m), w)
+def bitmatrix_decode(k, m, chunks, present, packetsize, w=8,
+                     technique=T_CAUCHY_ORIG_JERASURE):
+    bm = bitmatrix(_bit_coding_matrix(technique, k, m), w)
     pres = np.asarray(present, dtype=np.uint8)
     r = _ref.ecref_bitmatrix_decode(
         k, m, w, bm.ctypes.data_as(ctypes.c_void_p), _ptr_array(chunks),

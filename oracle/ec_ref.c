@@ -676,6 +676,70 @@ int ecref_matrix_cauchy_orig_jerasure(uint8_t *coding, int k, int m)
   return 0;
 }
 
+/* cauchy.c cauchy_n_ones, w=8: ones in the companion bitmatrix of e,
+ * i.e. sum over c of popcount(e * 2^c) (column c of the block is the bit
+ * pattern of e*2^c — ecref_matrix_to_bitmatrix below). jerasure computes
+ * the same value with an incremental recurrence; this is the direct form. */
+int ecref_cauchy_n_ones(uint8_t e)
+{
+  ecref_gf_init();
+  int no = 0;
+  uint8_t v = e;
+  for (int c = 0; c < 8; c++) {
+    no += __builtin_popcount(v);
+    v = ecref_gf_mul(v, 2);
+  }
+  return no;
+}
+
+/* cauchy.c cauchy_improve_coding_matrix: (1) scale each column so row 0
+ * becomes all ones; (2) for each later row, divide by the first element
+ * that strictly minimises the row's total bitmatrix ones. */
+void ecref_cauchy_improve_matrix(uint8_t *coding, int k, int m)
+{
+  ecref_gf_init();
+  for (int j = 0; j < k; j++) {
+    if (coding[j] != 1) {
+      uint8_t tmp = ecref_gf_inv(coding[j]);
+      for (int i = 1; i < m; i++)
+        coding[(size_t)i * k + j] =
+            ecref_gf_mul(coding[(size_t)i * k + j], tmp);
+      coding[j] = 1;
+    }
+  }
+  for (int i = 1; i < m; i++) {
+    uint8_t *row = coding + (size_t)i * k;
+    int bno = 0, bno_index = -1;
+    for (int j = 0; j < k; j++) bno += ecref_cauchy_n_ones(row[j]);
+    for (int j = 0; j < k; j++) {
+      if (row[j] == 1) continue;
+      uint8_t tmp = ecref_gf_inv(row[j]);
+      int tno = 0;
+      for (int col = 0; col < k; col++)
+        tno += ecref_cauchy_n_ones(ecref_gf_mul(row[col], tmp));
+      if (tno < bno) { bno = tno; bno_index = j; }
+    }
+    if (bno_index != -1) {
+      uint8_t tmp = ecref_gf_inv(row[bno_index]);
+      for (int j = 0; j < k; j++) row[j] = ecref_gf_mul(row[j], tmp);
+    }
+  }
+}
+
+/* cauchy.c cauchy_good_general_coding_matrix, general branch (m != 2):
+ * cauchy_original + improve. m == 2 would read jerasure's precomputed
+ * cbest tables (unsourceable in this container) => -EDOM. */
+int ecref_matrix_cauchy_good_jerasure(uint8_t *coding, int k, int m)
+{
+  if (m == 2)
+    return -EDOM;
+  int r = ecref_matrix_cauchy_orig_jerasure(coding, k, m);
+  if (r)
+    return r;
+  ecref_cauchy_improve_matrix(coding, k, m);
+  return 0;
+}
+
 /* jerasure.c jerasure_matrix_to_bitmatrix: block (i,j) column c holds the
  * bit pattern of coeff * 2^c (companion-matrix representation): bit row r
  * of the block = bit r of gf_mul(coeff, 1<<c). */

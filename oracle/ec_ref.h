@@ -153,6 +153,14 @@ int  ecref_decode16(uint8_t *const *chunks, const uint8_t *present,
  * round-trip/property tests.
  */
 int  ecref_matrix_cauchy_orig_jerasure(uint8_t *coding /* m x k */, int k, int m);
+/* cauchy.c cauchy_n_ones (w=8): ones in the companion bitmatrix of e. */
+int  ecref_cauchy_n_ones(uint8_t e);
+/* cauchy.c cauchy_improve_coding_matrix over the m x k coding rows. */
+void ecref_cauchy_improve_matrix(uint8_t *coding, int k, int m);
+/* cauchy.c cauchy_good_general_coding_matrix, general branch:
+ * cauchy_original + improve; m == 2 (jerasure's cbest-table branch,
+ * unsourceable here) => -EDOM. */
+int  ecref_matrix_cauchy_good_jerasure(uint8_t *coding /* m x k */, int k, int m);
 /* bitmat: (m*w) x (k*w) entries, one byte per bit, row-major */
 void ecref_matrix_to_bitmatrix(const uint8_t *coding, int k, int m, int w,
                                uint8_t *bitmat);

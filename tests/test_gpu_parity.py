@@ -242,19 +242,19 @@ def test_concurrent_calls_one_ctx():
     assert not errors, errors
 
 
-def test_cauchy_orig_bitmatrix_vs_oracle():
-    """jerasure cauchy_orig (bitmatrix/packet layout) on the GPU: the
-    LDS-staged XOR kernel must match the oracle's packet semantics
-    bit-exactly, encode and decode (host path + device batch)."""
+@pytest.mark.parametrize("tech", ["cauchy_orig", "cauchy_good"])
+def test_cauchy_bitmatrix_vs_oracle(tech):
+    """jerasure cauchy_orig/cauchy_good (bitmatrix/packet layout) on the
+    GPU: the LDS-staged XOR kernel must match the oracle's packet
+    semantics bit-exactly, encode and decode (host path + device batch)."""
     k, m, p = 4, 3, 2048
-    tech = "cauchy_orig"
     C = 8 * p * 4  # 4 superwords
     rng = np.random.default_rng(0xCA)
     ctx = ceph_amd.EcContext(k, m, tech, device=0, packetsize=p)
     try:
         data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
         got = ctx.encode_chunks(data)
-        want = oracle.bitmatrix_encode(k, m, data, p)
+        want = oracle.bitmatrix_encode(k, m, data, p, technique=tech)
         for j in range(m):
             assert np.array_equal(got[j], want[j]), j
         # decode exhaustive over all patterns
@@ -271,12 +271,13 @@ def test_cauchy_orig_bitmatrix_vs_oracle():
         ctx.close()
 
 
-def test_cauchy_orig_batch_roundtrip():
+@pytest.mark.parametrize("tech", ["cauchy_orig", "cauchy_good"])
+def test_cauchy_bitmatrix_batch_roundtrip(tech):
     k, m, p = 7, 3, 2048
     n = k + m
     C = 8 * p * 8
     S = 8
-    ctx = ceph_amd.EcContext(k, m, "cauchy_orig", device=0, packetsize=p)
+    ctx = ceph_amd.EcContext(k, m, tech, device=0, packetsize=p)
     try:
         nbytes = S * n * C
         d = ctx.dbuf_alloc(nbytes)
@@ -288,7 +289,7 @@ def test_cauchy_orig_batch_roundtrip():
         # spot-check stripe 3 vs oracle
         st = ref[3 * n * C:(3 * n + n) * C]
         data = [st[i * C:(i + 1) * C].copy() for i in range(k)]
-        want = oracle.bitmatrix_encode(k, m, data, p)
+        want = oracle.bitmatrix_encode(k, m, data, p, technique=tech)
         for j in range(m):
             assert np.array_equal(st[(k + j) * C:(k + j + 1) * C], want[j])
         # decode round trip
@@ -303,6 +304,32 @@ def test_cauchy_orig_batch_roundtrip():
         out = np.zeros(nbytes, np.uint8)
         ctx.download(out, d)
         assert np.array_equal(out, ref)
+    finally:
+        ctx.close()
+
+
+@pytest.mark.parametrize("tech", ["cauchy_orig", "cauchy_good"])
+def test_bitmatrix_delta_equals_reencode(tech):
+    """Bitmatrix parity-delta (schedule_apply_delta semantics,
+    ErasureCodeJerasure.cc:348-377): delta-apply per (data, coding) pair
+    must equal a full re-encode with the new data chunk."""
+    k, m, p = 5, 3, 2048
+    C = 8 * p * 2
+    rng = np.random.default_rng(0xDE17A)
+    ctx = ceph_amd.EcContext(k, m, tech, device=0, packetsize=p)
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        par = ctx.encode_chunks(data)
+        newc = rng.integers(0, 256, C, dtype=np.uint8)
+        delta = ctx.encode_delta(data[2], newc)
+        assert np.array_equal(delta, data[2] ^ newc)
+        par2 = [q.copy() for q in par]
+        for j in range(m):
+            ctx.apply_delta(delta, 2, k + j, par2[j])
+        data[2] = newc
+        want = oracle.bitmatrix_encode(k, m, data, p, technique=tech)
+        for j in range(m):
+            assert np.array_equal(par2[j], want[j]), j
     finally:
         ctx.close()
 
