@@ -68,8 +68,9 @@ def parity_selfcheck(ctx, dptr, args, seed, sample_stripes=2):
         exp = expected_fill(int(s) * stripe_bytes, k * C, seed)
         if not np.array_equal(host[:k * C], exp):
             raise AssertionError(f"stripe {s}: data region != expected fill")
-        if args.technique == "cauchy_orig":
-            want = oracle.bitmatrix_encode(k, m, data, 2048)
+        if args.technique in ("cauchy_orig", "cauchy_good"):
+            want = oracle.bitmatrix_encode(k, m, data, 2048,
+                                           technique=args.technique)
         elif args.technique == "jerasure_reed_sol_van_w16":
             want = oracle.encode_w16(k, m, data)
         else:

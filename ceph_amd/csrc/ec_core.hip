@@ -469,15 +469,48 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   uint8_t* obase = obuf + (long)blockIdx.y * cps * chunk_bytes;
   const long sw_off = sw * (long)w * pkt + (long)win * q;
 
-  for (int t = threadIdx.x; t < n_src * w * vq; t += blockDim.x) {
-    const int jc = t >> vq_shift;
-    const int v = t - (jc << vq_shift);
-    const int j = jc / w, c = jc - j * w;
-    const v4u* src = reinterpret_cast<const v4u*>(
-        sbase + (long)bp->src_ids[j] * chunk_bytes + sw_off + (long)c * pkt +
-        (long)v * 16);
-    const v4u d = NT ? __builtin_nontemporal_load(src) : *src;
-    *reinterpret_cast<v4u*>(s_data + (size_t)jc * q + (size_t)v * 16) = d;
+  // Window staging via LDS-DMA (global_load_lds_dwordx4): the LDS image
+  // is lane-linear in the item index (byte offset = t*16), exactly the
+  // wave-uniform-base + lane*16 layout the instruction writes. Unlike the
+  // v1 register-staged loop (1 outstanding load per thread -> the chip
+  // sat latency-starved: PMC SQ_WAIT_ANY/WAVE 0.79, LDS 4% active), every
+  // load of the phase is in flight at once and there is no ds_write pass;
+  // __syncthreads() drains the DMA (its fence emits vmcnt(0) while a
+  // glds is pending). aux=2 (nt) on the NT path: the window is read once
+  // by exactly one workgroup — no reuse to cache.
+  const int total_items = n_src * w * vq;
+  if ((total_items & 63) == 0) {
+    const int lane = threadIdx.x & 63;
+    const int nwaves = blockDim.x >> 6;
+    for (int t0 = (int)(threadIdx.x >> 6) * 64; t0 < total_items;
+         t0 += nwaves * 64) {
+      const int t = t0 + lane;
+      const int jc = t >> vq_shift;
+      const int v = t - (jc << vq_shift);
+      const int j = jc / w, c = jc - j * w;
+      const uint8_t* src = sbase + (long)bp->src_ids[j] * chunk_bytes +
+                           sw_off + (long)c * pkt + (long)v * 16;
+      auto gsrc = (const __attribute__((address_space(1))) uint32_t*)src;
+      auto ldst =
+          (__attribute__((address_space(3))) uint32_t*)(s_data +
+                                                        (size_t)t0 * 16);
+      if (NT)
+        __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 2);
+      else
+        __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);
+    }
+  } else {
+    // odd shapes (q < 128 with small k): register staging as in v1
+    for (int t = threadIdx.x; t < total_items; t += blockDim.x) {
+      const int jc = t >> vq_shift;
+      const int v = t - (jc << vq_shift);
+      const int j = jc / w, c = jc - j * w;
+      const v4u* src = reinterpret_cast<const v4u*>(
+          sbase + (long)bp->src_ids[j] * chunk_bytes + sw_off +
+          (long)c * pkt + (long)v * 16);
+      const v4u d = NT ? __builtin_nontemporal_load(src) : *src;
+      *reinterpret_cast<v4u*>(s_data + (size_t)jc * q + (size_t)v * 16) = d;
+    }
   }
   __syncthreads();
 
