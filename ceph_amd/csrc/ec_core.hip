@@ -450,9 +450,9 @@ template <bool NT, bool ACCUM = false>
 __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
-    int windows_per_sw) {
+    int windows_per_sw, int wpb, long n_windows) {
   const EcBitParams* bp = (const EcBitParams*)blob;
-  const uint16_t* g_ops = (const uint16_t*)(blob + sizeof(EcBitParams));
+  const uint32_t* g_ops = (const uint32_t*)(blob + sizeof(EcBitParams));
   extern __shared__ uint8_t smem[];
   const int n_src = bp->n_src, w = bp->w, pkt = bp->pkt, q = bp->q;
   const int vq = q >> 4, vq_shift = bp->vq_shift;
@@ -460,84 +460,97 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   uint8_t* s_data = smem;                       // n_src*w*q bytes
   uint16_t* s_ops = (uint16_t*)(smem + (size_t)n_src * w * q);
   const int n_ops = bp->row_off[n_rows];
-  for (int t = threadIdx.x; t < n_ops; t += blockDim.x) s_ops[t] = g_ops[t];
+  // ops staged once per block (dword copies; host pads the blob), then
+  // reused across all wpb windows — v1/v2 re-read and re-wrote the blob
+  // per window (one window per block)
+  for (int t = threadIdx.x; t < (n_ops + 1) / 2; t += blockDim.x)
+    reinterpret_cast<uint32_t*>(s_ops)[t] = g_ops[t];
 
-  const long tile = blockIdx.x;
-  const int win = (int)(tile % windows_per_sw);
-  const long sw = tile / windows_per_sw;
   const uint8_t* sbase = buf + (long)blockIdx.y * cps * chunk_bytes;
   uint8_t* obase = obuf + (long)blockIdx.y * cps * chunk_bytes;
-  const long sw_off = sw * (long)w * pkt + (long)win * q;
-
-  // Window staging via LDS-DMA (global_load_lds_dwordx4): the LDS image
-  // is lane-linear in the item index (byte offset = t*16), exactly the
-  // wave-uniform-base + lane*16 layout the instruction writes. Unlike the
-  // v1 register-staged loop (1 outstanding load per thread -> the chip
-  // sat latency-starved: PMC SQ_WAIT_ANY/WAVE 0.79, LDS 4% active), every
-  // load of the phase is in flight at once and there is no ds_write pass;
-  // __syncthreads() drains the DMA (its fence emits vmcnt(0) while a
-  // glds is pending). aux=2 (nt) on the NT path: the window is read once
-  // by exactly one workgroup — no reuse to cache.
   const int total_items = n_src * w * vq;
-  if ((total_items & 63) == 0) {
-    const int lane = threadIdx.x & 63;
-    const int nwaves = blockDim.x >> 6;
-    for (int t0 = (int)(threadIdx.x >> 6) * 64; t0 < total_items;
-         t0 += nwaves * 64) {
-      const int t = t0 + lane;
-      const int jc = t >> vq_shift;
-      const int v = t - (jc << vq_shift);
-      const int j = jc / w, c = jc - j * w;
-      const uint8_t* src = sbase + (long)bp->src_ids[j] * chunk_bytes +
-                           sw_off + (long)c * pkt + (long)v * 16;
-      auto gsrc = (const __attribute__((address_space(1))) uint32_t*)src;
-      auto ldst =
-          (__attribute__((address_space(3))) uint32_t*)(s_data +
-                                                        (size_t)t0 * 16);
-      if (NT)
-        __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 2);
-      else
-        __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);
-    }
-  } else {
-    // odd shapes (q < 128 with small k): register staging as in v1
-    for (int t = threadIdx.x; t < total_items; t += blockDim.x) {
-      const int jc = t >> vq_shift;
-      const int v = t - (jc << vq_shift);
-      const int j = jc / w, c = jc - j * w;
-      const v4u* src = reinterpret_cast<const v4u*>(
-          sbase + (long)bp->src_ids[j] * chunk_bytes + sw_off +
-          (long)c * pkt + (long)v * 16);
-      const v4u d = NT ? __builtin_nontemporal_load(src) : *src;
-      *reinterpret_cast<v4u*>(s_data + (size_t)jc * q + (size_t)v * 16) = d;
-    }
-  }
-  __syncthreads();
+  const long w_begin = (long)blockIdx.x * wpb;
+  const long w_end = w_begin + wpb < n_windows ? w_begin + wpb : n_windows;
 
-  // Item-parallel compute (one 16B vec of one output row per item): A/B
-  // showed this beats a row-per-wave readlane variant — with q=512 the
-  // row-per-wave form idles half of each wave (vq=32) and larger q
-  // collapses residency; LDS op reads broadcast cheaply.
-  for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
-    const int r = t >> vq_shift;
-    const int v = t - (r << vq_shift);
-    v4u* dst = reinterpret_cast<v4u*>(
-        obase + (long)bp->out_ids[r / w] * chunk_bytes + sw * (long)w * pkt +
-        (long)(r % w) * pkt + (long)win * q + (long)v * 16);
-    // ACCUM = parity-delta apply (schedule_apply_delta semantics,
-    // ErasureCodeJerasure.cc:348-377): XOR into the existing parity
-    v4u acc = ACCUM ? *dst : v4u{0, 0, 0, 0};
-    const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
-    for (int o = b0; o < b1; o++) {
-      const int jc = s_ops[o];
-      const v4u d = *reinterpret_cast<const v4u*>(
-          s_data + (size_t)jc * q + (size_t)v * 16);
-      acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
+  for (long wt = w_begin; wt < w_end; wt++) {
+    const int win = (int)(wt % windows_per_sw);
+    const long sw = wt / windows_per_sw;
+    const long sw_off = sw * (long)w * pkt + (long)win * q;
+    // all waves done computing the previous window before its LDS image
+    // is overwritten (first window: orders the ops staging)
+    __syncthreads();
+
+    // Window staging via LDS-DMA (global_load_lds_dwordx4): the LDS
+    // image is lane-linear in the item index (byte offset = t*16),
+    // exactly the wave-uniform-base + lane*16 layout the instruction
+    // writes. Unlike the v1 register-staged loop (1 outstanding load per
+    // thread -> the chip sat latency-starved: PMC SQ_WAIT_ANY/WAVE 0.79,
+    // LDS 4% active), every load of the phase is in flight at once and
+    // there is no ds_write pass; __syncthreads() drains the DMA (its
+    // fence emits vmcnt(0) while a glds is pending). aux=2 (nt) on the
+    // NT path: each window is read once by exactly one workgroup.
+    if ((total_items & 63) == 0) {
+      const int lane = threadIdx.x & 63;
+      const int nwaves = blockDim.x >> 6;
+      for (int t0 = (int)(threadIdx.x >> 6) * 64; t0 < total_items;
+           t0 += nwaves * 64) {
+        const int t = t0 + lane;
+        const int jc = t >> vq_shift;
+        const int v = t - (jc << vq_shift);
+        const int j = jc / w, c = jc - j * w;
+        const uint8_t* src = sbase + (long)bp->src_ids[j] * chunk_bytes +
+                             sw_off + (long)c * pkt + (long)v * 16;
+        auto gsrc = (const __attribute__((address_space(1))) uint32_t*)src;
+        auto ldst =
+            (__attribute__((address_space(3))) uint32_t*)(s_data +
+                                                          (size_t)t0 * 16);
+        if (NT)
+          __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 2);
+        else
+          __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);
+      }
+    } else {
+      // odd shapes (q < 128 with small k): register staging as in v1
+      for (int t = threadIdx.x; t < total_items; t += blockDim.x) {
+        const int jc = t >> vq_shift;
+        const int v = t - (jc << vq_shift);
+        const int j = jc / w, c = jc - j * w;
+        const v4u* src = reinterpret_cast<const v4u*>(
+            sbase + (long)bp->src_ids[j] * chunk_bytes + sw_off +
+            (long)c * pkt + (long)v * 16);
+        const v4u d = NT ? __builtin_nontemporal_load(src) : *src;
+        *reinterpret_cast<v4u*>(s_data + (size_t)jc * q + (size_t)v * 16) =
+            d;
+      }
     }
-    if (NT)
-      __builtin_nontemporal_store(acc, dst);
-    else
-      *dst = acc;
+    __syncthreads();
+
+    // Item-parallel compute (one 16B vec of one output row per item): A/B
+    // showed this beats a row-per-wave readlane variant — with q=512 the
+    // row-per-wave form idles half of each wave (vq=32) and larger q
+    // collapses residency; LDS op reads broadcast cheaply.
+    for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
+      const int r = t >> vq_shift;
+      const int v = t - (r << vq_shift);
+      v4u* dst = reinterpret_cast<v4u*>(
+          obase + (long)bp->out_ids[r / w] * chunk_bytes +
+          sw * (long)w * pkt + (long)(r % w) * pkt + (long)win * q +
+          (long)v * 16);
+      // ACCUM = parity-delta apply (schedule_apply_delta semantics,
+      // ErasureCodeJerasure.cc:348-377): XOR into the existing parity
+      v4u acc = ACCUM ? *dst : v4u{0, 0, 0, 0};
+      const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
+      for (int o = b0; o < b1; o++) {
+        const int jc = s_ops[o];
+        const v4u d = *reinterpret_cast<const v4u*>(
+            s_data + (size_t)jc * q + (size_t)v * 16);
+        acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
+      }
+      if (NT)
+        __builtin_nontemporal_store(acc, dst);
+      else
+        *dst = acc;
+    }
   }
 }
 
@@ -1264,6 +1277,7 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   }
   hdr.row_off[n_rows] = (uint16_t)ops.size();
   if (ops.size() > 0xffff) return ECX_ERR_INVAL;
+  if (ops.size() & 1) ops.push_back(0);  // pad: kernel copies ops as dwords
   size_t blob = sizeof(EcBitParams) + ops.size() * 2;
   int r = ensure_jobs(ctx, s, blob);
   if (r != ECX_OK) return r;
@@ -1276,7 +1290,19 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
 
   const long sw_per_chunk = (long)(chunk_bytes / ((size_t)w * pkt));
   const int windows_per_sw = pkt / q;
-  dim3 grid((unsigned)(sw_per_chunk * windows_per_sw), (unsigned)n_stripes);
+  const long n_windows = sw_per_chunk * windows_per_sw;
+  // windows per block: amortizes the ops staging and block start/drain
+  // over several LDS-window rounds without growing the LDS footprint
+  // (q stays small => 8 blocks/CU residency); MI355X sweep default 8.
+  static const int env_wpb = [] {
+    const char* v = getenv("ECX_BITW");
+    long n = v ? atol(v) : 8;
+    if (n < 1) n = 1;
+    if (n > 4096) n = 4096;
+    return (int)n;
+  }();
+  const int wpb = env_wpb;
+  dim3 grid((unsigned)((n_windows + wpb - 1) / wpb), (unsigned)n_stripes);
   size_t lds = (size_t)n_src * w * q + ((ops.size() * 2 + 15) & ~15ull);
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
   auto kfn = env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
@@ -1285,7 +1311,7 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
                              : ec_bitmatrix_kernel<false, false>);
   hipLaunchKernelGGL(kfn, grid, dim3(256), lds, s.stream, d_buf, d_obuf,
                      s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
-                     windows_per_sw);
+                     windows_per_sw, wpb, n_windows);
   HIP_TRY(hipGetLastError());
   if (time_it) {
     HIP_TRY(hipEventRecord(s.ev_stop, s.stream));

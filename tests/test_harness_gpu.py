@@ -134,14 +134,19 @@ def test_lrc_decode_exhaustive_gpu():
      "partialread,partialwrite,zeroinout,paritydelta,crcencodedecode,"
      "directreads"),
     (["-P", "technique=cauchy_orig", "-P", "k=4", "-P", "m=2"],
-     "partialread,partialwrite,zeroinout,directreads"),
+     "partialread,partialwrite,zeroinout,paritydelta,directreads"),
+    (["-P", "technique=cauchy_good", "-P", "k=4", "-P", "m=3"],
+     "partialread,partialwrite,zeroinout,paritydelta,crcencodedecode,"
+     "directreads"),
 ])
 def test_mi355x_flags_mirror_reference(profile, expected):
     """Per-technique optimization flags mirror the owning reference
     plugin exactly (ErasureCodeIsa.h:66-79, ErasureCodeJerasure.h:52-63):
     isa techniques claim OPTIMIZED always and CRC for reed_sol_van /
     cauchy-at-m=1; jerasure techniques claim OPTIMIZED only for
-    reed_sol_van and CRC for all but reed_sol_van and cauchy_orig."""
+    reed_sol_van and CRC for all but reed_sol_van and cauchy_orig;
+    paritydelta is claimed for every technique (bitmatrix deltas are
+    implemented via the schedule-delta kernel)."""
     r = run_bench("-p", "mi355x", *profile, "--flags")
     assert r.returncode == 0, r.stderr
     assert r.stdout.strip() == expected
