@@ -554,6 +554,135 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   }
 }
 
+// Software-pipelined bitmatrix variant: two LDS window buffers; each wave
+// issues its glds batch for window t+1, then waits ONLY for window t's
+// batch (counted s_waitcnt vmcnt(G) — vmcnt retires in order) and
+// computes t while t+1 streams in. Costs 2x data LDS (half the block
+// residency) but overlaps the load latency the barrier otherwise
+// serialises; raw s_barrier + lgkmcnt-only waits so the in-flight glds
+// survives the barrier (hipcc's __syncthreads would drain vmcnt(0)).
+#define ECX_WAITV(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
+__device__ __forceinline__ void ecx_wait_vmcnt(int g) {
+  switch (g) {
+    case 0: ECX_WAITV(0); break;
+    case 1: ECX_WAITV(1); break;
+    case 2: ECX_WAITV(2); break;
+    case 3: ECX_WAITV(3); break;
+    case 4: ECX_WAITV(4); break;
+    case 5: ECX_WAITV(5); break;
+    case 6: ECX_WAITV(6); break;
+    case 7: ECX_WAITV(7); break;
+    case 8: ECX_WAITV(8); break;
+    case 9: ECX_WAITV(9); break;
+    case 10: ECX_WAITV(10); break;
+    case 11: ECX_WAITV(11); break;
+    case 12: ECX_WAITV(12); break;
+    case 13: ECX_WAITV(13); break;
+    case 14: ECX_WAITV(14); break;
+    case 15: ECX_WAITV(15); break;
+    default: ECX_WAITV(16); break;
+  }
+}
+__device__ __forceinline__ void ecx_raw_barrier() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+}
+
+template <bool NT, bool ACCUM = false>
+__global__ __launch_bounds__(256, 2) void ec_bitmatrix_pipe_kernel(
+    const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
+    const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
+    int windows_per_sw, int wpb, long n_windows) {
+  const EcBitParams* bp = (const EcBitParams*)blob;
+  const uint32_t* g_ops = (const uint32_t*)(blob + sizeof(EcBitParams));
+  extern __shared__ uint8_t smem[];
+  const int n_src = bp->n_src, w = bp->w, pkt = bp->pkt, q = bp->q;
+  const int vq = q >> 4, vq_shift = bp->vq_shift;
+  const int n_rows = bp->n_out * w;
+  const size_t buf_bytes = (size_t)n_src * w * q;
+  uint8_t* s_buf0 = smem;
+  uint8_t* s_buf1 = smem + buf_bytes;
+  uint16_t* s_ops = (uint16_t*)(smem + 2 * buf_bytes);
+  const int n_ops = bp->row_off[n_rows];
+  for (int t = threadIdx.x; t < (n_ops + 1) / 2; t += blockDim.x)
+    reinterpret_cast<uint32_t*>(s_ops)[t] = g_ops[t];
+
+  const uint8_t* sbase = buf + (long)blockIdx.y * cps * chunk_bytes;
+  uint8_t* obase = obuf + (long)blockIdx.y * cps * chunk_bytes;
+  const int total_items = n_src * w * vq;  // host guarantees %64 == 0
+  const int chunks = total_items >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wave = (int)(threadIdx.x >> 6);
+  const int nwaves = (int)(blockDim.x >> 6);
+  // uniform per-wave glds count: tail waves re-issue a wrapped chunk
+  // (same source -> same LDS bytes; racing identical writes are benign)
+  const int G = (chunks + nwaves - 1) / nwaves;
+  const long w_begin = (long)blockIdx.x * wpb;
+  const long w_end = w_begin + wpb < n_windows ? w_begin + wpb : n_windows;
+  if (w_begin >= w_end) return;
+
+  auto issue = [&](long wt, uint8_t* dstbuf) {
+    const int win = (int)(wt % windows_per_sw);
+    const long sw = wt / windows_per_sw;
+    const long sw_off = sw * (long)w * pkt + (long)win * q;
+    for (int g = 0; g < G; g++) {
+      int chunk = wave + g * nwaves;
+      if (chunk >= chunks) chunk -= chunks;  // wrap (duplicate, benign)
+      const int t0 = chunk << 6;
+      const int t = t0 + lane;
+      const int jc = t >> vq_shift;
+      const int v = t - (jc << vq_shift);
+      const int j = jc / w, c = jc - j * w;
+      const uint8_t* src = sbase + (long)bp->src_ids[j] * chunk_bytes +
+                           sw_off + (long)c * pkt + (long)v * 16;
+      auto gsrc = (const __attribute__((address_space(1))) uint32_t*)src;
+      auto ldst = (__attribute__((address_space(3))) uint32_t*)(dstbuf +
+                                                               (size_t)t0 *
+                                                                   16);
+      if (NT)
+        __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 2);
+      else
+        __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);
+    }
+  };
+
+  issue(w_begin, s_buf0);
+  for (long wt = w_begin; wt < w_end; wt++) {
+    uint8_t* cur = (wt - w_begin) & 1 ? s_buf1 : s_buf0;
+    if (wt + 1 < w_end) {
+      issue(wt + 1, (wt + 1 - w_begin) & 1 ? s_buf1 : s_buf0);
+      ecx_wait_vmcnt(G);  // window wt landed; wt+1 still streaming
+    } else {
+      ecx_wait_vmcnt(0);
+    }
+    ecx_raw_barrier();  // every wave's window-wt DMA landed; s_ops staged
+
+    const int win = (int)(wt % windows_per_sw);
+    const long sw = wt / windows_per_sw;
+    for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
+      const int r = t >> vq_shift;
+      const int v = t - (r << vq_shift);
+      v4u* dst = reinterpret_cast<v4u*>(
+          obase + (long)bp->out_ids[r / w] * chunk_bytes +
+          sw * (long)w * pkt + (long)(r % w) * pkt + (long)win * q +
+          (long)v * 16);
+      v4u acc = ACCUM ? *dst : v4u{0, 0, 0, 0};
+      const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
+      for (int o = b0; o < b1; o++) {
+        const int jc = s_ops[o];
+        const v4u d = *reinterpret_cast<const v4u*>(
+            cur + (size_t)jc * q + (size_t)v * 16);
+        acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
+      }
+      if (NT)
+        __builtin_nontemporal_store(acc, dst);
+      else
+        *dst = acc;
+    }
+    ecx_raw_barrier();  // all reads of `cur` done before wt+2 overwrites it
+  }
+}
+
 // delta = a ^ b (encode_delta; replaces galois_region_xor / xor_gen).
 __global__ __launch_bounds__(256) void ec_xor_kernel(
     const uint8_t* __restrict__ a, const uint8_t* __restrict__ b,
@@ -1302,13 +1431,27 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
     return (int)n;
   }();
   const int wpb = env_wpb;
+  static const int env_pipe = [] {
+    const char* v = getenv("ECX_BITPIPE");
+    return v ? atoi(v) : 0;
+  }();
   dim3 grid((unsigned)((n_windows + wpb - 1) / wpb), (unsigned)n_stripes);
-  size_t lds = (size_t)n_src * w * q + ((ops.size() * 2 + 15) & ~15ull);
+  const size_t data_bytes = (size_t)n_src * w * q;
+  const size_t ops_bytes = (ops.size() * 2 + 15) & ~15ull;
+  const int total_items = n_src * w * (q >> 4);
+  const bool pipe = env_pipe && wpb > 1 && (total_items & 63) == 0 &&
+                    2 * data_bytes + ops_bytes <= 160 * 1024;
+  size_t lds = (pipe ? 2 * data_bytes : data_bytes) + ops_bytes;
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
-  auto kfn = env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
-                             : ec_bitmatrix_kernel<true, false>)
-                    : (accum ? ec_bitmatrix_kernel<false, true>
-                             : ec_bitmatrix_kernel<false, false>);
+  auto kfn =
+      pipe ? (env_nt ? (accum ? ec_bitmatrix_pipe_kernel<true, true>
+                              : ec_bitmatrix_pipe_kernel<true, false>)
+                     : (accum ? ec_bitmatrix_pipe_kernel<false, true>
+                              : ec_bitmatrix_pipe_kernel<false, false>))
+           : (env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
+                              : ec_bitmatrix_kernel<true, false>)
+                     : (accum ? ec_bitmatrix_kernel<false, true>
+                              : ec_bitmatrix_kernel<false, false>));
   hipLaunchKernelGGL(kfn, grid, dim3(256), lds, s.stream, d_buf, d_obuf,
                      s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
                      windows_per_sw, wpb, n_windows);

@@ -302,6 +302,14 @@ def chunk_size(technique, k, stripe_width, w=8):
 
 # ---- fast CPU-baseline batch ops (AVX2/OpenMP; kind="port" in bench) ----
 
+def cpu_first_touch(batch):
+    """NUMA-spread the batch's pages (parallel first-touch) BEFORE filling
+    content — a single-thread numpy fill places every page on one node
+    and the OpenMP encode then starves the other socket(s)."""
+    _cpu.eccpu_first_touch(batch.ctypes.data_as(ctypes.c_void_p),
+                           ctypes.c_size_t(batch.nbytes))
+
+
 def cpu_threads():
     return _cpu.eccpu_threads()
 

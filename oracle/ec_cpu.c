@@ -144,6 +144,28 @@ static void encode_block(int k, int m, const uint8_t *tables /* m*k*32 */,
   }
 }
 
+/* Parallel first-touch of a batch buffer: page placement follows the
+ * first writer (Linux first-touch NUMA policy), so a buffer filled by a
+ * single numpy thread lands on ONE node and the OpenMP encode then
+ * starves every other socket's cores (measured on a GPU-box host: 101
+ * GiB/s at 32 threads vs 46 at 64 with single-thread fill). Touch pages
+ * with the same static partitioning the encode loops use BEFORE filling
+ * content; later writes do not move pages. */
+int eccpu_first_touch(uint8_t *base, size_t bytes)
+{
+  const size_t blk = 2u << 20; /* 2 MiB */
+  long nblk = (long)((bytes + blk - 1) / blk);
+#ifdef _OPENMP
+#pragma omp parallel for schedule(static)
+#endif
+  for (long b = 0; b < nblk; b++) {
+    size_t off = (size_t)b * blk;
+    size_t n = bytes - off < blk ? bytes - off : blk;
+    memset(base + off, 0, n);
+  }
+  return 0;
+}
+
 int eccpu_threads(void)
 {
 #ifdef _OPENMP

@@ -49,6 +49,8 @@ def mem_bw_probe():
 def run_cell(stripes, threads, do_decode, secs=6.0):
     set_threads(threads)
     batch = np.empty(stripes * (K + M) * C, dtype=np.uint8)
+    if os.environ.get("RECON_TOUCH", "1") == "1":
+        oracle.cpu_first_touch(batch)
     batch[:] = np.frombuffer(os.urandom(1 << 20), np.uint8).repeat(
         (batch.nbytes + (1 << 20) - 1) // (1 << 20))[:batch.nbytes]
     present = np.ones(K + M, np.uint8)
