@@ -450,7 +450,17 @@ template <bool NT, bool ACCUM = false>
 __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
-    int windows_per_sw, int wpb, long n_windows) {
+    int windows_per_sw, int wpb, long n_windows, int stagger) {
+  // Optional start-phase stagger (knob ECX_BITSTAGGER): co-resident
+  // blocks otherwise run their DMA-wait / compute phases in lockstep
+  // (dispatched together, identical per-window timing), leaving HBM idle
+  // during every block's compute segment — PMC shows 73% wave-parked
+  // with LDS/VALU both < 20% busy. A one-time per-block phase offset
+  // breaks the convoy; offsets persist because the cycle time is uniform.
+  if (stagger) {
+    const int ph = (int)(blockIdx.x + blockIdx.y) & 7;
+    for (int i = 0; i < ph * stagger; i++) __builtin_amdgcn_s_sleep(8);
+  }
   const EcBitParams* bp = (const EcBitParams*)blob;
   const uint32_t* g_ops = (const uint32_t*)(blob + sizeof(EcBitParams));
   extern __shared__ uint8_t smem[];
@@ -1432,7 +1442,23 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   }();
   const int wpb = env_wpb;
   static const int env_pipe = [] {
+    // 2-buffer glds pipeline: measured SLOWER (18.0 vs 10.9 ms at wpb=8,
+    // profiles r2) — halved residency outweighs the in-block overlap at
+    // 8-blocks/CU occupancy, as the CDNA guide's regime note predicts.
+    // Kept behind the knob as a recorded negative.
     const char* v = getenv("ECX_BITPIPE");
+    return v ? atoi(v) : 0;
+  }();
+  static const int env_bt = [] {
+    // block size: 128-thread blocks double the independent blocks per CU
+    // (16 at q=128) so co-resident blocks interleave DMA-wait and
+    // compute phases more finely
+    const char* v = getenv("ECX_BITT");
+    int n = v ? atoi(v) : 256;
+    return (n == 64 || n == 128 || n == 256) ? n : 256;
+  }();
+  static const int env_stagger = [] {
+    const char* v = getenv("ECX_BITSTAGGER");
     return v ? atoi(v) : 0;
   }();
   dim3 grid((unsigned)((n_windows + wpb - 1) / wpb), (unsigned)n_stripes);
@@ -1443,18 +1469,23 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
                     2 * data_bytes + ops_bytes <= 160 * 1024;
   size_t lds = (pipe ? 2 * data_bytes : data_bytes) + ops_bytes;
   if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
-  auto kfn =
-      pipe ? (env_nt ? (accum ? ec_bitmatrix_pipe_kernel<true, true>
-                              : ec_bitmatrix_pipe_kernel<true, false>)
-                     : (accum ? ec_bitmatrix_pipe_kernel<false, true>
-                              : ec_bitmatrix_pipe_kernel<false, false>))
-           : (env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
-                              : ec_bitmatrix_kernel<true, false>)
-                     : (accum ? ec_bitmatrix_kernel<false, true>
-                              : ec_bitmatrix_kernel<false, false>));
-  hipLaunchKernelGGL(kfn, grid, dim3(256), lds, s.stream, d_buf, d_obuf,
-                     s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
-                     windows_per_sw, wpb, n_windows);
+  if (pipe) {
+    auto kfn = env_nt ? (accum ? ec_bitmatrix_pipe_kernel<true, true>
+                               : ec_bitmatrix_pipe_kernel<true, false>)
+                      : (accum ? ec_bitmatrix_pipe_kernel<false, true>
+                               : ec_bitmatrix_pipe_kernel<false, false>);
+    hipLaunchKernelGGL(kfn, grid, dim3(256), lds, s.stream, d_buf, d_obuf,
+                       s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
+                       windows_per_sw, wpb, n_windows);
+  } else {
+    auto kfn = env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
+                               : ec_bitmatrix_kernel<true, false>)
+                      : (accum ? ec_bitmatrix_kernel<false, true>
+                               : ec_bitmatrix_kernel<false, false>);
+    hipLaunchKernelGGL(kfn, grid, dim3(env_bt), lds, s.stream, d_buf,
+                       d_obuf, s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
+                       windows_per_sw, wpb, n_windows, env_stagger);
+  }
   HIP_TRY(hipGetLastError());
   if (time_it) {
     HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
