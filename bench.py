@@ -69,7 +69,7 @@ def parity_selfcheck(ctx, dptr, args, seed, sample_stripes=2):
         if not np.array_equal(host[:k * C], exp):
             raise AssertionError(f"stripe {s}: data region != expected fill")
         if args.technique in ("cauchy_orig", "cauchy_good"):
-            want = oracle.bitmatrix_encode(k, m, data, 2048,
+            want = oracle.bitmatrix_encode(k, m, data, args.packetsize,
                                            technique=args.technique)
         elif args.technique == "jerasure_reed_sol_van_w16":
             want = oracle.encode_w16(k, m, data)
@@ -518,6 +518,8 @@ def main():
     ap.add_argument("--k", type=int, default=8)
     ap.add_argument("--m", type=int, default=3)
     ap.add_argument("--technique", default="reed_sol_van")
+    ap.add_argument("--packetsize", type=int, default=2048,
+                    help="bitmatrix techniques only (jerasure packetsize)")
     ap.add_argument("--erasures", type=int, default=3)
     ap.add_argument("--seed", type=lambda x: int(x, 0), default=0xEC)
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -581,7 +583,8 @@ def main():
     seed = args.seed + rank
 
     ctx = ceph_amd.EcContext(k, m, args.technique, device=device,
-                             n_streams=args.streams)
+                             n_streams=args.streams,
+                             packetsize=args.packetsize)
     dptr = ctx.dbuf_alloc(buf_bytes)
     ctx.fill_random(dptr, buf_bytes, seed)
     ctx.sync()

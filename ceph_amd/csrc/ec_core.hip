@@ -450,7 +450,7 @@ template <bool NT, bool ACCUM = false>
 __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
-    int windows_per_sw, int wpb, long n_windows, int stagger) {
+    int windows_per_sw, int wpb, long n_windows, int stagger, int xcdmap) {
   // Optional start-phase stagger (knob ECX_BITSTAGGER): co-resident
   // blocks otherwise run their DMA-wait / compute phases in lockstep
   // (dispatched together, identical per-window timing), leaving HBM idle
@@ -480,11 +480,31 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   uint8_t* obase = obuf + (long)blockIdx.y * cps * chunk_bytes;
   const int total_items = n_src * w * vq;
   const long w_begin = (long)blockIdx.x * wpb;
-  const long w_end = w_begin + wpb < n_windows ? w_begin + wpb : n_windows;
 
-  for (long wt = w_begin; wt < w_end; wt++) {
-    const int win = (int)(wt % windows_per_sw);
-    const long sw = wt / windows_per_sw;
+  for (long it = 0; it < wpb; it++) {
+    int win;
+    long sw;
+    if (xcdmap) {
+      // XCD-cooperative mapping (knob ECX_BITXCD; host enables only when
+      // windows_per_sw == 8 and the grid divides cleanly): consecutive
+      // blocks land on XCDs b % 8, so blocks b..b+63 — dispatched
+      // together — cover 8 superwords x 8 window slots with each XCD's 8
+      // blocks reading the SAME superword at different 256 B offsets.
+      // Their requests arrive temporally clustered, so DRAM sees whole
+      // 2 KiB rows instead of 1-of-8 row churn (membench: strided-granule
+      // reads are the gap between this kernel and the copy ceiling).
+      const int x = (int)(blockIdx.x & 7);
+      const int slot = (int)((blockIdx.x >> 3) & 7);
+      const long tg = (long)(blockIdx.x >> 6);
+      sw = (tg * wpb + it) * 8 + x;
+      win = slot;
+      if (sw >= (n_windows >> 3)) break;
+    } else {
+      const long wt = w_begin + it;
+      if (wt >= n_windows) break;
+      win = (int)(wt % windows_per_sw);
+      sw = wt / windows_per_sw;
+    }
     const long sw_off = sw * (long)w * pkt + (long)win * q;
     // all waves done computing the previous window before its LDS image
     // is overwritten (first window: orders the ops staging)
@@ -1461,6 +1481,10 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
     const char* v = getenv("ECX_BITSTAGGER");
     return v ? atoi(v) : 0;
   }();
+  static const int env_xcd = [] {
+    const char* v = getenv("ECX_BITXCD");
+    return v ? atoi(v) : 0;
+  }();
   dim3 grid((unsigned)((n_windows + wpb - 1) / wpb), (unsigned)n_stripes);
   const size_t data_bytes = (size_t)n_src * w * q;
   const size_t ops_bytes = (ops.size() * 2 + 15) & ~15ull;
@@ -1478,13 +1502,17 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
                        s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
                        windows_per_sw, wpb, n_windows);
   } else {
+    const long n_blocks = (n_windows + wpb - 1) / wpb;
+    const int xcdmap =
+        env_xcd && windows_per_sw == 8 && n_windows % ((long)wpb * 64) == 0 &&
+        n_blocks % 64 == 0;
     auto kfn = env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
                                : ec_bitmatrix_kernel<true, false>)
                       : (accum ? ec_bitmatrix_kernel<false, true>
                                : ec_bitmatrix_kernel<false, false>);
     hipLaunchKernelGGL(kfn, grid, dim3(env_bt), lds, s.stream, d_buf,
                        d_obuf, s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
-                       windows_per_sw, wpb, n_windows, env_stagger);
+                       windows_per_sw, wpb, n_windows, env_stagger, xcdmap);
   }
   HIP_TRY(hipGetLastError());
   if (time_it) {
