@@ -41,6 +41,82 @@ __global__ __launch_bounds__(256) void read_pattern(
     sink[threadIdx.x] = acc.x;  // never true for random fill
 }
 
+// ---- differential ladder: from pure reads to the full bitmatrix shape.
+// MODE 0: glds staging + barrier + LDS-gather compute, result XOR-sunk
+//         (no global stores)
+// MODE 1: MODE 0 + nontemporal 16 B stores in the real scattered pattern
+// MODE 2: MODE 1 but plain (temporal) stores
+// Shape fixed at k=8 w=8 q=256 pkt=2048 m=3 (the bench default); ops
+// table synthesised on-device with the same per-row count (32).
+template <int MODE>
+__global__ __launch_bounds__(256, 2) void stage_pattern(
+    const uint8_t* __restrict__ bufbase, long n_sw, int wpb,
+    uint32_t* __restrict__ sink) {
+  const uint8_t* buf = bufbase + (size_t)blockIdx.y * (11u << 20);
+  uint8_t* obuf = const_cast<uint8_t*>(buf) + (8u << 20);
+  constexpr int K = 8, W = 8, Q = 256, PKT = 2048, M = 3;
+  constexpr int VQ = Q / 16, ITEMS = K * W * VQ, ROWS = M * W;
+  __shared__ uint8_t s_data[K * W * Q];
+  __shared__ uint16_t s_ops[ROWS * 32];
+  for (int t = threadIdx.x; t < ROWS * 32; t += blockDim.x)
+    s_ops[t] = (uint16_t)(((t >> 5) * 7 + (t & 31) * 13 + ((t & 31) >> 2)) %
+               (K * W));
+  const long chunk_bytes = 1 << 20;
+  v4u acc_sink = {0, 0, 0, 0};
+  for (long it = 0; it < wpb; it++) {
+    const long wt = (long)blockIdx.x * wpb + it;
+    if (wt >= n_sw * (PKT / Q)) break;
+    const int win = (int)(wt % (PKT / Q));
+    const long sw = wt / (PKT / Q);
+    const long sw_off = sw * (long)W * PKT + (long)win * Q;
+    __syncthreads();
+    {
+      const int lane = threadIdx.x & 63;
+      const int nwaves = blockDim.x >> 6;
+      for (int t0 = (int)(threadIdx.x >> 6) * 64; t0 < ITEMS;
+           t0 += nwaves * 64) {
+        const int t = t0 + lane;
+        const int jc = t >> 4;
+        const int v = t & 15;
+        const int j = jc >> 3, c = jc & 7;
+        const uint8_t* src = buf + (long)j * chunk_bytes + sw_off +
+                             (long)c * PKT + (long)v * 16;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) uint32_t*)src,
+            (__attribute__((address_space(3))) uint32_t*)(s_data +
+                                                          (size_t)t0 * 16),
+            16, 0, 2);
+      }
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < ROWS * VQ; t += blockDim.x) {
+      const int r = t >> 4;
+      const int v = t & 15;
+      v4u acc = {0, 0, 0, 0};
+      for (int o = 0; o < 32; o++) {
+        const int jc = s_ops[r * 32 + o];
+        const v4u d = *reinterpret_cast<const v4u*>(
+            s_data + (size_t)jc * Q + (size_t)v * 16);
+        acc.x ^= d.x; acc.y ^= d.y; acc.z ^= d.z; acc.w ^= d.w;
+      }
+      if (MODE == 0) {
+        acc_sink.x ^= acc.x; acc_sink.y ^= acc.y;
+        acc_sink.z ^= acc.z; acc_sink.w ^= acc.w;
+      } else {
+        v4u* dst = reinterpret_cast<v4u*>(
+            obuf + (long)(r >> 3) * chunk_bytes + sw * (long)W * PKT +
+            (long)(r & 7) * PKT + (long)win * Q + (long)v * 16);
+        if (MODE == 1)
+          __builtin_nontemporal_store(acc, dst);
+        else
+          *dst = acc;
+      }
+    }
+  }
+  if (acc_sink.x + acc_sink.y + acc_sink.z + acc_sink.w == 0xdeadbeefu)
+    sink[threadIdx.x] = acc_sink.x;
+}
+
 int main() {
   const size_t total = 8ull << 30;  // 8 GiB working set
   uint8_t* d;
@@ -90,6 +166,47 @@ int main() {
            "\"read_GBs\": %.0f}\n",
            p.name, p.gran, p.stride, 3 * bytes / (ms * 1e6));
     fflush(stdout);
+  }
+  // ---- ladder runs: 1 MiB chunks, 64 superwords/chunk, many stripes
+  {
+    const long n_sw = 64;       // one chunk's superwords (1 MiB chunks)
+    const long stripes = 512;   // 11 MiB arena per stripe within 8 GiB
+    const int wpb = 8;
+    const long windows = n_sw * 8;
+    const unsigned blocks_per_stripe = (unsigned)(windows / wpb);
+    const dim3 grid(blocks_per_stripe, (unsigned)stripes);
+    // per launch: read 16 KiB + write 6 KiB per window
+    const double rd = (double)stripes * windows * 16384;
+    const double wr = (double)stripes * windows * 6144;
+    for (int mode = 0; mode < 3; mode++) {
+      HT(hipDeviceSynchronize());
+      HT(hipEventRecord(e0));
+      for (int r = 0; r < 3; r++) {
+        switch (mode) {
+          case 0:
+            hipLaunchKernelGGL(stage_pattern<0>, grid, dim3(256), 0, 0, d,
+                               n_sw, wpb, sink);
+            break;
+          case 1:
+            hipLaunchKernelGGL(stage_pattern<1>, grid, dim3(256), 0, 0, d,
+                               n_sw, wpb, sink);
+            break;
+          case 2:
+            hipLaunchKernelGGL(stage_pattern<2>, grid, dim3(256), 0, 0, d,
+                               n_sw, wpb, sink);
+            break;
+        }
+      }
+      HT(hipEventRecord(e1));
+      HT(hipEventSynchronize(e1));
+      float ms = 0;
+      HT(hipEventElapsedTime(&ms, e0, e1));
+      const double bytes = mode == 0 ? rd : rd + wr;
+      printf("{\"ladder_mode\": %d, \"GBs\": %.0f, \"read_GBs_equiv\":"
+             " %.0f}\n", mode, 3 * bytes / (ms * 1e6),
+             3 * rd / (ms * 1e6));
+      fflush(stdout);
+    }
   }
   return 0;
 }
