@@ -46,6 +46,11 @@ __global__ __launch_bounds__(256) void read_pattern(
 //         (no global stores)
 // MODE 1: MODE 0 + nontemporal 16 B stores in the real scattered pattern
 // MODE 2: MODE 1 but plain (temporal) stores
+// MODE 3: MODE 1 volume, but stores to a CONTIGUOUS per-block region
+//         (granularity probe: same bytes, ideal locality)
+// MODE 4: output rows accumulated in LDS across the superword's 8
+//         windows, flushed as contiguous 2 KiB packets (the candidate
+//         kernel design)
 // Shape fixed at k=8 w=8 q=256 pkt=2048 m=3 (the bench default); ops
 // table synthesised on-device with the same per-row count (32).
 template <int MODE>
@@ -57,6 +62,7 @@ __global__ __launch_bounds__(256, 2) void stage_pattern(
   constexpr int K = 8, W = 8, Q = 256, PKT = 2048, M = 3;
   constexpr int VQ = Q / 16, ITEMS = K * W * VQ, ROWS = M * W;
   __shared__ uint8_t s_data[K * W * Q];
+  __shared__ uint8_t s_out[MODE == 4 ? ROWS * PKT : 16];
   __shared__ uint16_t s_ops[ROWS * 32];
   for (int t = threadIdx.x; t < ROWS * 32; t += blockDim.x)
     s_ops[t] = (uint16_t)(((t >> 5) * 7 + (t & 31) * 13 + ((t & 31) >> 2)) %
@@ -102,6 +108,13 @@ __global__ __launch_bounds__(256, 2) void stage_pattern(
       if (MODE == 0) {
         acc_sink.x ^= acc.x; acc_sink.y ^= acc.y;
         acc_sink.z ^= acc.z; acc_sink.w ^= acc.w;
+      } else if (MODE == 3) {
+        v4u* dst = reinterpret_cast<v4u*>(
+            obuf + ((long)blockIdx.x * wpb + it) * (ROWS * Q) + (long)t * 16);
+        __builtin_nontemporal_store(acc, dst);
+      } else if (MODE == 4) {
+        *reinterpret_cast<v4u*>(s_out + (size_t)r * PKT + (size_t)win * Q +
+                                (size_t)v * 16) = acc;
       } else {
         v4u* dst = reinterpret_cast<v4u*>(
             obuf + (long)(r >> 3) * chunk_bytes + sw * (long)W * PKT +
@@ -110,6 +123,20 @@ __global__ __launch_bounds__(256, 2) void stage_pattern(
           __builtin_nontemporal_store(acc, dst);
         else
           *dst = acc;
+      }
+    }
+    if (MODE == 4 && win == (PKT / Q) - 1) {
+      // superword complete: flush all rows as contiguous 2 KiB packets
+      __syncthreads();
+      for (int t = threadIdx.x; t < ROWS * (PKT / 16); t += blockDim.x) {
+        const int r = t / (PKT / 16);
+        const int v = t - r * (PKT / 16);
+        const v4u val = *reinterpret_cast<const v4u*>(
+            s_out + (size_t)r * PKT + (size_t)v * 16);
+        __builtin_nontemporal_store(
+            val, reinterpret_cast<v4u*>(obuf + (long)(r >> 3) * chunk_bytes +
+                                        sw * (long)W * PKT +
+                                        (long)(r & 7) * PKT + (long)v * 16));
       }
     }
   }
@@ -178,7 +205,7 @@ int main() {
     // per launch: read 16 KiB + write 6 KiB per window
     const double rd = (double)stripes * windows * 16384;
     const double wr = (double)stripes * windows * 6144;
-    for (int mode = 0; mode < 3; mode++) {
+    for (int mode = 0; mode < 5; mode++) {
       HT(hipDeviceSynchronize());
       HT(hipEventRecord(e0));
       for (int r = 0; r < 3; r++) {
@@ -193,6 +220,14 @@ int main() {
             break;
           case 2:
             hipLaunchKernelGGL(stage_pattern<2>, grid, dim3(256), 0, 0, d,
+                               n_sw, wpb, sink);
+            break;
+          case 3:
+            hipLaunchKernelGGL(stage_pattern<3>, grid, dim3(256), 0, 0, d,
+                               n_sw, wpb, sink);
+            break;
+          case 4:
+            hipLaunchKernelGGL(stage_pattern<4>, grid, dim3(256), 0, 0, d,
                                n_sw, wpb, sink);
             break;
         }
