@@ -443,10 +443,20 @@ struct EcBitParams {
   int src_ids[ECX_MAX_K];
   int out_ids[ECX_MAX_OUT];
   uint16_t row_off[ECX_MAX_OUT * 8 + 1];  // prefix offsets into ops[]
+  // rows are stored sorted by op count so the 4 rows sharing a wave have
+  // similar lengths (divergence waste 5.6-7.8% -> 0.6-3.5% measured);
+  // row_map[r] = original row id (output chunk out_ids[orig/w], packet
+  // orig%w)
+  uint8_t row_map[ECX_MAX_OUT * 8];
   // blob continues with uint16 ops[row_off[n_rows]]: values j*w+c
 };
 
-template <bool NT, bool ACCUM = false>
+// VQS >= 0 specialises the window geometry at compile time (w = 8,
+// q = 16 << VQS): the generic form's runtime divisions/shifts and
+// loop bounds cost ~15% against the measured ladder probe of the same
+// structure (tools/membench.hip stage_pattern: static 4.94 TB/s vs the
+// generic kernel's 4.23 at the same HBM pattern).
+template <bool NT, bool ACCUM = false, int VQS = -1>
 __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
@@ -464,8 +474,10 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   const EcBitParams* bp = (const EcBitParams*)blob;
   const uint32_t* g_ops = (const uint32_t*)(blob + sizeof(EcBitParams));
   extern __shared__ uint8_t smem[];
-  const int n_src = bp->n_src, w = bp->w, pkt = bp->pkt, q = bp->q;
-  const int vq = q >> 4, vq_shift = bp->vq_shift;
+  const int n_src = bp->n_src, pkt = bp->pkt;
+  const int w = VQS >= 0 ? 8 : bp->w;
+  const int q = VQS >= 0 ? (16 << VQS) : bp->q;
+  const int vq = q >> 4, vq_shift = VQS >= 0 ? VQS : bp->vq_shift;
   const int n_rows = bp->n_out * w;
   uint8_t* s_data = smem;                       // n_src*w*q bytes
   uint16_t* s_ops = (uint16_t*)(smem + (size_t)n_src * w * q);
@@ -562,9 +574,10 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
     for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
       const int r = t >> vq_shift;
       const int v = t - (r << vq_shift);
+      const int orig = bp->row_map[r];
       v4u* dst = reinterpret_cast<v4u*>(
-          obase + (long)bp->out_ids[r / w] * chunk_bytes +
-          sw * (long)w * pkt + (long)(r % w) * pkt + (long)win * q +
+          obase + (long)bp->out_ids[orig / w] * chunk_bytes +
+          sw * (long)w * pkt + (long)(orig % w) * pkt + (long)win * q +
           (long)v * 16);
       // ACCUM = parity-delta apply (schedule_apply_delta semantics,
       // ErasureCodeJerasure.cc:348-377): XOR into the existing parity
@@ -692,9 +705,10 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_pipe_kernel(
     for (int t = threadIdx.x; t < n_rows * vq; t += blockDim.x) {
       const int r = t >> vq_shift;
       const int v = t - (r << vq_shift);
+      const int orig = bp->row_map[r];
       v4u* dst = reinterpret_cast<v4u*>(
-          obase + (long)bp->out_ids[r / w] * chunk_bytes +
-          sw * (long)w * pkt + (long)(r % w) * pkt + (long)win * q +
+          obase + (long)bp->out_ids[orig / w] * chunk_bytes +
+          sw * (long)w * pkt + (long)(orig % w) * pkt + (long)win * q +
           (long)v * 16);
       v4u acc = ACCUM ? *dst : v4u{0, 0, 0, 0};
       const int b0 = bp->row_off[r], b1 = bp->row_off[r + 1];
@@ -1428,8 +1442,20 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   for (int j = 0; j < n_out; j++) hdr.out_ids[j] = out_ids[j];
   std::vector<uint16_t> ops;
   ops.reserve((size_t)n_rows * W / 2);
+  // sort rows by op count (waves span several rows; uniform lengths kill
+  // the max-over-rows divergence in the compute loop)
+  std::vector<std::pair<int, int>> order(n_rows);
   for (int r = 0; r < n_rows; r++) {
-    hdr.row_off[r] = (uint16_t)ops.size();
+    int cnt = 0;
+    const uint8_t* row = bit_rows + (size_t)r * W;
+    for (int c = 0; c < W; c++) cnt += row[c] != 0;
+    order[r] = {cnt, r};
+  }
+  std::sort(order.begin(), order.end());
+  for (int rr = 0; rr < n_rows; rr++) {
+    const int r = order[rr].second;
+    hdr.row_map[rr] = (uint8_t)r;
+    hdr.row_off[rr] = (uint16_t)ops.size();
     const uint8_t* row = bit_rows + (size_t)r * W;
     for (int c = 0; c < W; c++)
       if (row[c]) ops.push_back((uint16_t)c);
@@ -1506,10 +1532,16 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
     const int xcdmap =
         env_xcd && windows_per_sw == 8 && n_windows % ((long)wpb * 64) == 0 &&
         n_blocks % 64 == 0;
-    auto kfn = env_nt ? (accum ? ec_bitmatrix_kernel<true, true>
-                               : ec_bitmatrix_kernel<true, false>)
-                      : (accum ? ec_bitmatrix_kernel<false, true>
-                               : ec_bitmatrix_kernel<false, false>);
+#define ECX_BMK(VQS_)                                                \
+  (env_nt ? (accum ? ec_bitmatrix_kernel<true, true, VQS_>           \
+                   : ec_bitmatrix_kernel<true, false, VQS_>)         \
+          : (accum ? ec_bitmatrix_kernel<false, true, VQS_>          \
+                   : ec_bitmatrix_kernel<false, false, VQS_>))
+    auto kfn = ECX_BMK(-1);
+    if (w == 8 && vq_shift == 3) kfn = ECX_BMK(3);
+    if (w == 8 && vq_shift == 4) kfn = ECX_BMK(4);
+    if (w == 8 && vq_shift == 5) kfn = ECX_BMK(5);
+#undef ECX_BMK
     hipLaunchKernelGGL(kfn, grid, dim3(env_bt), lds, s.stream, d_buf,
                        d_obuf, s.d_jobs, (long)chunk_bytes, ctx->k + ctx->m,
                        windows_per_sw, wpb, n_windows, env_stagger, xcdmap);
@@ -1738,7 +1770,7 @@ int ecx_decode_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
     return run_bitmatrix(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr,
                          plan.survivors.data(), ctx->k, plan.erased.data(),
                          (int)plan.erased.size(), plan.rows.data(),
-                         n_stripes, chunk_bytes, false);
+                         n_stripes, chunk_bytes, /*time_it=*/true);
   }
   if (ctx->is_w16()) {
     Decode16Plan plan;
@@ -1748,7 +1780,7 @@ int ecx_decode_batch(ecx_ctx* ctx, void* dptr, long n_stripes,
     return run_matmul16(ctx, slot, (const uint8_t*)dptr, (uint8_t*)dptr,
                         plan.survivors.data(), ctx->k, plan.erased.data(),
                         (int)plan.erased.size(), plan.rows.data(),
-                        n_stripes, chunk_bytes, false, false);
+                        n_stripes, chunk_bytes, false, /*time_it=*/true);
   }
   DecodePlan plan;
   int r = get_decode_plan(ctx, present_mask, plan);
