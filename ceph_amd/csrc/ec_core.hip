@@ -762,6 +762,10 @@ __global__ __launch_bounds__(256) void ec_fill_random_kernel(
 // Host-side: context, streams, decode-row LRU, launches
 // ---------------------------------------------------------------------------
 
+// bounded query-spin waits (defined in the host-pointer section below)
+static hipError_t wait_event(hipEvent_t ev);
+static hipError_t wait_stream(hipStream_t st);
+
 namespace {
 
 struct Slot {
@@ -792,6 +796,8 @@ struct Slot {
   hipEvent_t ev_pipe[2] = {nullptr, nullptr};
   double last_ms = -1.0;
   bool timed = false;
+  // captured single-tile host-call graphs, keyed by (tl, n_src, n_out)
+  std::map<uint64_t, hipGraphExec_t> graphs;
   std::recursive_mutex mu;
 };
 
@@ -951,6 +957,10 @@ void ecx_destroy(ecx_ctx* ctx) {
     if (s.d_pipe) (void)hipFree(s.d_pipe);
     if (s.h_pparams) (void)hipHostFree(s.h_pparams);
     if (s.d_pparams) (void)hipFree(s.d_pparams);
+    for (auto& [k, ge] : s.graphs) {
+      (void)k;
+      (void)hipGraphExecDestroy(ge);
+    }
     if (s.stream) (void)hipStreamDestroy(s.stream);
   }
   delete ctx;
@@ -1223,7 +1233,7 @@ static int launch_matmul(ecx_ctx* ctx, Slot& s, const uint8_t* d_buf,
   HIP_TRY(hipSetDevice(ctx->device));
 
   // wait for any in-flight param upload on this slot, then stage params
-  HIP_TRY(hipEventSynchronize(s.ev_param));
+  HIP_TRY(wait_event(s.ev_param));
   std::memcpy(s.h_params, &params, sizeof(EcLaunchParams));
   HIP_TRY(hipMemcpyAsync(s.d_params, s.h_params, sizeof(EcLaunchParams),
                          hipMemcpyHostToDevice, s.stream));
@@ -1273,7 +1283,7 @@ static int run_matmul(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
 static int ensure_jobs(ecx_ctx* ctx, Slot& s, size_t bytes) {
   if (s.jobs_bytes >= bytes) return ECX_OK;
   HIP_TRY(hipSetDevice(ctx->device));
-  HIP_TRY(hipEventSynchronize(s.ev_jobs));
+  HIP_TRY(wait_event(s.ev_jobs));
   if (s.h_jobs) (void)hipHostFree(s.h_jobs);
   if (s.d_jobs) (void)hipFree(s.d_jobs);
   s.h_jobs = nullptr;
@@ -1320,7 +1330,7 @@ static int run_slices(ecx_ctx* ctx, int slot_i, void* const* d_chunks,
   size_t blob = off_bsl + (size_t)n_blocks * 4;
   int r = ensure_jobs(ctx, s, blob);
   if (r != ECX_OK) return r;
-  HIP_TRY(hipEventSynchronize(s.ev_jobs));
+  HIP_TRY(wait_event(s.ev_jobs));
 
   uint64_t* ptrs = (uint64_t*)(s.h_jobs + off_ptrs);
   long* svecs = (long*)(s.h_jobs + off_vecs);
@@ -1357,7 +1367,7 @@ static int run_slices(ecx_ctx* ctx, int slot_i, void* const* d_chunks,
                   n_src);
     // per-slice NULL chunks are handled inside the kernel, not via cls
     fill_params(&p, f, src_ids, n_src, out_ids + j0, nj, sub.data(), nullptr);
-    HIP_TRY(hipEventSynchronize(s.ev_param));
+    HIP_TRY(wait_event(s.ev_param));
     std::memcpy(s.h_params, &p, sizeof(EcLaunchParams));
     HIP_TRY(hipMemcpyAsync(s.d_params, s.h_params, sizeof(EcLaunchParams),
                            hipMemcpyHostToDevice, s.stream));
@@ -1466,7 +1476,7 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   size_t blob = sizeof(EcBitParams) + ops.size() * 2;
   int r = ensure_jobs(ctx, s, blob);
   if (r != ECX_OK) return r;
-  HIP_TRY(hipEventSynchronize(s.ev_jobs));
+  HIP_TRY(wait_event(s.ev_jobs));
   std::memcpy(s.h_jobs, &hdr, sizeof(hdr));
   std::memcpy(s.h_jobs + sizeof(hdr), ops.data(), ops.size() * 2);
   HIP_TRY(hipMemcpyAsync(s.d_jobs, s.h_jobs, blob, hipMemcpyHostToDevice,
@@ -1631,7 +1641,7 @@ static int run_matmul16(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
     size_t blob = sizeof(EcLaunch16);
     int r = ensure_jobs(ctx, s, blob);
     if (r != ECX_OK) return r;
-    HIP_TRY(hipEventSynchronize(s.ev_jobs));
+    HIP_TRY(wait_event(s.ev_jobs));
     std::memcpy(s.h_jobs, &p, blob);
     HIP_TRY(hipMemcpyAsync(s.d_jobs, s.h_jobs, blob, hipMemcpyHostToDevice,
                            s.stream));
@@ -1954,7 +1964,7 @@ int ecx_matmul_chunks_host(ecx_ctx* ctx, const uint8_t* const* srcs,
     HIP_TRY(hipMemcpyAsync(outs[j], s.d_stage + (size_t)(n_src + j) * bytes,
                            bytes, hipMemcpyDeviceToHost, s.stream));
   }
-  HIP_TRY(hipStreamSynchronize(s.stream));
+  HIP_TRY(wait_stream(s.stream));
   return ECX_OK;
 }
 
@@ -2023,6 +2033,49 @@ int ecx_last_kernel_ms(ecx_ctx* ctx, int slot, double* ms) {
 }  // extern "C"
 
 // ---- host-pointer (plugin) path ----
+
+// Blocking hipEventSynchronize/hipStreamSynchronize park the thread in
+// the driver (~37 us each measured on MI355X, r1 profiles) — ruinous for
+// sub-ms host calls (the OSD's real call shape is many small calls,
+// ECUtil.cc:485-514). A bounded hipEventQuery/hipStreamQuery spin costs
+// ~1-2 us when the work is already done or finishes soon; fall back to
+// the blocking wait after ~2 ms so large calls still park politely.
+// ECX_SPINWAIT=0 disables.
+static int env_spinwait() {
+  static const int v = [] {
+    const char* e = getenv("ECX_SPINWAIT");
+    return e ? atoi(e) : 1;
+  }();
+  return v;
+}
+
+static hipError_t wait_event(hipEvent_t ev) {
+  if (env_spinwait()) {
+    const auto t0 = std::chrono::steady_clock::now();
+    for (;;) {
+      hipError_t e = hipEventQuery(ev);
+      if (e != hipErrorNotReady) return e;
+      if (std::chrono::steady_clock::now() - t0 >
+          std::chrono::milliseconds(2))
+        break;
+    }
+  }
+  return hipEventSynchronize(ev);
+}
+
+static hipError_t wait_stream(hipStream_t st) {
+  if (env_spinwait()) {
+    const auto t0 = std::chrono::steady_clock::now();
+    for (;;) {
+      hipError_t e = hipStreamQuery(st);
+      if (e != hipErrorNotReady) return e;
+      if (std::chrono::steady_clock::now() - t0 >
+          std::chrono::milliseconds(2))
+        break;
+    }
+  }
+  return hipStreamSynchronize(st);
+}
 
 static int ensure_stage(ecx_ctx* ctx, Slot& s, size_t bytes) {
   if (s.stage_bytes >= bytes) return ECX_OK;
@@ -2224,8 +2277,8 @@ static void par_copy(const std::vector<CopyOp>& ops) {
 static int ensure_pipe(ecx_ctx* ctx, Slot& s, size_t bytes) {
   if (s.pipe_bytes >= bytes) return ECX_OK;
   HIP_TRY(hipSetDevice(ctx->device));
-  HIP_TRY(hipEventSynchronize(s.ev_pipe[0]));
-  HIP_TRY(hipEventSynchronize(s.ev_pipe[1]));
+  HIP_TRY(wait_event(s.ev_pipe[0]));
+  HIP_TRY(wait_event(s.ev_pipe[1]));
   if (s.h_pipe) (void)hipHostFree(s.h_pipe);
   if (s.d_pipe) (void)hipFree(s.d_pipe);
   s.h_pipe = nullptr;
@@ -2280,8 +2333,8 @@ static int staged_host_call(ecx_ctx* ctx, Slot& s,
   int r = ensure_pipe(ctx, s, (size_t)n_slots * chunk_bytes);
   if (r != ECX_OK) return r;
   HIP_TRY(hipSetDevice(ctx->device));
-  HIP_TRY(hipEventSynchronize(s.ev_pipe[0]));
-  HIP_TRY(hipEventSynchronize(s.ev_pipe[1]));
+  HIP_TRY(wait_event(s.ev_pipe[0]));
+  HIP_TRY(wait_event(s.ev_pipe[1]));
   std::vector<CopyOp> ops;
   int lo = n_slots, hi = 0;
   for (int i = 0; i < n_src; i++) {
@@ -2310,7 +2363,7 @@ static int staged_host_call(ecx_ctx* ctx, Slot& s,
                            s.d_pipe + (size_t)out_slot[j] * chunk_bytes,
                            chunk_bytes, hipMemcpyDeviceToHost, s.stream));
   }
-  HIP_TRY(hipStreamSynchronize(s.stream));
+  HIP_TRY(wait_stream(s.stream));
   ops.clear();
   for (int j = 0; j < n_out; j++)
     if (hout[j])
@@ -2360,8 +2413,8 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
     const ecx::GF8& f = ecx::gf8();
     // a previous successful call drained both pipe events, but an errored
     // one may not have: make the pinned param area provably safe to rewrite
-    HIP_TRY(hipEventSynchronize(s.ev_pipe[0]));
-    HIP_TRY(hipEventSynchronize(s.ev_pipe[1]));
+    HIP_TRY(wait_event(s.ev_pipe[0]));
+    HIP_TRY(wait_event(s.ev_pipe[1]));
     s.pparams_kind = 0;
     for (int g = 0; g < groups; g++) {
       const int j0 = 4 * g, nj = std::min(4, n_out - j0);
@@ -2378,6 +2431,74 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
   }
 
   const long T = (long)((chunk_bytes + TS - 1) / TS);
+
+  // Single-tile small-call fast path: capture the (H2D, kernels, D2H)
+  // chain once per (tl, n_src, n_out) into a hipGraph and replay it —
+  // one submit (~10-16 us replay floor) instead of 3-5 driver calls.
+  // Addresses are stable (pinned h_pipe / d_pipe / resident d_pparams),
+  // and the captured work is content-independent, so the graph stays
+  // valid across calls and across param-table rewrites. ECX_HGRAPH=0
+  // disables. (DESIGN: the OSD call shape is many small calls.)
+  static const int env_hg = [] {
+    const char* e = getenv("ECX_HGRAPH");
+    return e ? atoi(e) : 1;
+  }();
+  if (T == 1 && env_hg) {
+    const size_t tl = chunk_bytes;
+    uint8_t* hbuf = s.h_pipe;
+    uint8_t* dbuf = s.d_pipe;
+    HIP_TRY(wait_event(s.ev_pipe[0]));
+    HIP_TRY(wait_event(s.ev_pipe[1]));
+    std::vector<CopyOp> gops;
+    for (int i = 0; i < n_src; i++)
+      if (srcs[i]) gops.push_back({hbuf + (size_t)i * tl, srcs[i], tl});
+    par_copy(gops);
+    const uint64_t key =
+        (uint64_t)tl | ((uint64_t)n_src << 44) | ((uint64_t)n_out << 52);
+    auto it = s.graphs.find(key);
+    if (it == s.graphs.end()) {
+      HIP_TRY(
+          hipStreamBeginCapture(s.stream, hipStreamCaptureModeThreadLocal));
+      int rc = ECX_OK;
+      hipError_t he = hipMemcpyAsync(dbuf, hbuf, (size_t)n_src * tl,
+                                     hipMemcpyHostToDevice, s.stream);
+      if (he != hipSuccess) rc = map_hip(he);
+      const MatmulCfg c = matmul_cfg((long)(tl >> 4), 1);
+      for (int g = 0; g < groups && rc == ECX_OK; g++) {
+        const int nj = std::min(4, n_out - 4 * g);
+        rc = matmul_dispatch(s.stream, dbuf, dbuf, s.d_pparams + g, nj,
+                             n_src, false, c, tl, cps);
+      }
+      if (rc == ECX_OK) {
+        he = hipMemcpyAsync(hbuf + (size_t)n_src * tl,
+                            dbuf + (size_t)n_src * tl, (size_t)n_out * tl,
+                            hipMemcpyDeviceToHost, s.stream);
+        if (he != hipSuccess) rc = map_hip(he);
+      }
+      hipGraph_t g = nullptr;
+      he = hipStreamEndCapture(s.stream, &g);
+      if (he != hipSuccess) return map_hip(he);
+      if (rc != ECX_OK) {
+        if (g) (void)hipGraphDestroy(g);
+        return rc;
+      }
+      hipGraphExec_t ge = nullptr;
+      he = hipGraphInstantiate(&ge, g, nullptr, nullptr, 0);
+      (void)hipGraphDestroy(g);
+      if (he != hipSuccess) return map_hip(he);
+      it = s.graphs.emplace(key, ge).first;
+    }
+    HIP_TRY(hipGraphLaunch(it->second, s.stream));
+    HIP_TRY(hipEventRecord(s.ev_pipe[0], s.stream));
+    HIP_TRY(wait_event(s.ev_pipe[0]));
+    gops.clear();
+    for (int j = 0; j < n_out; j++)
+      if (outs[j])
+        gops.push_back({outs[j], hbuf + ((size_t)n_src + j) * tl, tl});
+    par_copy(gops);
+    return ECX_OK;
+  }
+
   size_t tl_of[2] = {0, 0}, off_of[2] = {0, 0};
   std::vector<CopyOp> ops;
   ops.reserve(cps);
@@ -2389,7 +2510,7 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
     uint8_t* dbuf = s.d_pipe + (size_t)b * cps * TS;
     // buffer b holds tile t-2 until its D2H lands; wait, scatter it out,
     // then refill — the CPU work here overlaps tile t-1's DMA + kernels
-    HIP_TRY(hipEventSynchronize(s.ev_pipe[b]));
+    HIP_TRY(wait_event(s.ev_pipe[b]));
     if (t >= 2) {
       ops.clear();
       for (int j = 0; j < n_out; j++)
@@ -2429,7 +2550,7 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
   // drain the last one or two in-flight tiles
   for (long t = std::max(0L, T - 2); t < T; t++) {
     const int b = (int)(t & 1);
-    HIP_TRY(hipEventSynchronize(s.ev_pipe[b]));
+    HIP_TRY(wait_event(s.ev_pipe[b]));
     uint8_t* hbuf = s.h_pipe + (size_t)b * cps * TS;
     ops.clear();
     for (int j = 0; j < n_out; j++)
@@ -2537,7 +2658,7 @@ int ecx_encode_chunks_host(ecx_ctx* ctx, const uint8_t* const* data,
     HIP_TRY(hipMemcpyAsync(parity[j], s.d_stage + (size_t)(k + j) * chunk_bytes,
                            chunk_bytes, hipMemcpyDeviceToHost, s.stream));
   }
-  HIP_TRY(hipStreamSynchronize(s.stream));
+  HIP_TRY(wait_stream(s.stream));
   return ECX_OK;
 }
 
@@ -2594,7 +2715,7 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
       HIP_TRY(hipMemcpyAsync(chunks[e], s.d_stage + (size_t)e * chunk_bytes,
                              chunk_bytes, hipMemcpyDeviceToHost, s.stream));
     }
-    HIP_TRY(hipStreamSynchronize(s.stream));
+    HIP_TRY(wait_stream(s.stream));
     return ECX_OK;
   }
 
@@ -2658,7 +2779,7 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
       HIP_TRY(hipMemcpyAsync(chunks[e], s.d_stage + (size_t)e * chunk_bytes,
                              chunk_bytes, hipMemcpyDeviceToHost, s.stream));
     }
-    HIP_TRY(hipStreamSynchronize(s.stream));
+    HIP_TRY(wait_stream(s.stream));
     return ECX_OK;
   }
 
@@ -2715,7 +2836,7 @@ int ecx_decode_chunks_host(ecx_ctx* ctx, uint8_t* const* chunks,
     HIP_TRY(hipMemcpyAsync(chunks[e], s.d_stage + (size_t)e * chunk_bytes,
                            chunk_bytes, hipMemcpyDeviceToHost, s.stream));
   }
-  HIP_TRY(hipStreamSynchronize(s.stream));
+  HIP_TRY(wait_stream(s.stream));
   return ECX_OK;
 }
 
@@ -2739,7 +2860,7 @@ int ecx_encode_delta_host(ecx_ctx* ctx, const uint8_t* old_data,
   if (r != ECX_OK) return r;
   HIP_TRY(hipMemcpyAsync(delta, s.d_stage + 2 * bytes, bytes,
                          hipMemcpyDeviceToHost, s.stream));
-  HIP_TRY(hipStreamSynchronize(s.stream));
+  HIP_TRY(wait_stream(s.stream));
   return ECX_OK;
 }
 
@@ -2761,7 +2882,7 @@ int ecx_apply_delta_host(ecx_ctx* ctx, const uint8_t* delta, int data_shard,
   if (r != ECX_OK) return r;
   HIP_TRY(hipMemcpyAsync(parity, s.d_stage + bytes, bytes,
                          hipMemcpyDeviceToHost, s.stream));
-  HIP_TRY(hipStreamSynchronize(s.stream));
+  HIP_TRY(wait_stream(s.stream));
   return ECX_OK;
 }
 
