@@ -2039,7 +2039,7 @@ int ecx_last_kernel_ms(ecx_ctx* ctx, int slot, double* ms) {
 // sub-ms host calls (the OSD's real call shape is many small calls,
 // ECUtil.cc:485-514). A bounded hipEventQuery/hipStreamQuery spin costs
 // ~1-2 us when the work is already done or finishes soon; fall back to
-// the blocking wait after ~2 ms so large calls still park politely.
+// the blocking wait after ~500 us so large calls still park politely.
 // ECX_SPINWAIT=0 disables.
 static int env_spinwait() {
   static const int v = [] {
@@ -2056,7 +2056,7 @@ static hipError_t wait_event(hipEvent_t ev) {
       hipError_t e = hipEventQuery(ev);
       if (e != hipErrorNotReady) return e;
       if (std::chrono::steady_clock::now() - t0 >
-          std::chrono::milliseconds(2))
+          std::chrono::microseconds(500))
         break;
     }
   }
@@ -2070,7 +2070,7 @@ static hipError_t wait_stream(hipStream_t st) {
       hipError_t e = hipStreamQuery(st);
       if (e != hipErrorNotReady) return e;
       if (std::chrono::steady_clock::now() - t0 >
-          std::chrono::milliseconds(2))
+          std::chrono::microseconds(500))
         break;
     }
   }
@@ -2279,6 +2279,13 @@ static int ensure_pipe(ecx_ctx* ctx, Slot& s, size_t bytes) {
   HIP_TRY(hipSetDevice(ctx->device));
   HIP_TRY(wait_event(s.ev_pipe[0]));
   HIP_TRY(wait_event(s.ev_pipe[1]));
+  // captured single-tile graphs bake in the h_pipe/d_pipe addresses:
+  // reallocating invalidates every one of them
+  for (auto& [k, ge] : s.graphs) {
+    (void)k;
+    (void)hipGraphExecDestroy(ge);
+  }
+  s.graphs.clear();
   if (s.h_pipe) (void)hipHostFree(s.h_pipe);
   if (s.d_pipe) (void)hipFree(s.d_pipe);
   s.h_pipe = nullptr;
