@@ -520,6 +520,8 @@ def main():
     ap.add_argument("--technique", default="reed_sol_van")
     ap.add_argument("--packetsize", type=int, default=2048,
                     help="bitmatrix techniques only (jerasure packetsize)")
+    ap.add_argument("--dry-run", action="store_true",
+                    help="emit the per-rank run plan and exit (no GPU)")
     ap.add_argument("--erasures", type=int, default=3)
     ap.add_argument("--seed", type=lambda x: int(x, 0), default=0xEC)
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -551,6 +553,23 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     dist = None
     device = local_rank
+    if args.dry_run:
+        # SCALE-run readiness check (CPU-only, no torch/GPU): emit the
+        # per-rank plan the 8-GPU driver launch would use so CI can
+        # verify arg/env parsing and the rank->device map without
+        # hardware.
+        k, m, C, S = args.k, args.m, args.chunk_bytes, args.stripes
+        print(json.dumps({
+            "dry_run": True, "rank": rank, "world": world,
+            "local_rank": local_rank, "device": local_rank,
+            "backend": "nccl" if world > 1 else None,
+            "config": args.config, "technique": args.technique,
+            "k": k, "m": m, "chunk_bytes": C, "stripes_per_gpu": S,
+            "buf_bytes": S * (k + m) * C, "steps": args.steps,
+            "warmup": args.warmup, "scaling": "weak",
+            "master_addr": os.environ.get("MASTER_ADDR"),
+            "master_port": os.environ.get("MASTER_PORT")}))
+        return
     if world > 1:
         import torch
         import torch.distributed as tdist

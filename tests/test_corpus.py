@@ -61,3 +61,14 @@ def test_gpu_plugin_matches_committed_corpus(dirname, tmp_path):
     shutil.copytree(os.path.join(CORPUS, dirname), tmp_path / renamed)
     r = run_check(tmp_path, "mi355x", renamed)
     assert r.returncode == 0, (dirname, r.stderr)
+
+
+def test_corpus_check_script():
+    """tools/corpus_check.sh (the one-command external parity pin,
+    INTEGRATION.md) replays the whole committed corpus clean with the
+    CPU oracle plugin."""
+    r = subprocess.run(
+        [os.path.join(ROOT, "tools", "corpus_check.sh"), CORPUS, "oracle"],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "0 failed" in r.stdout
