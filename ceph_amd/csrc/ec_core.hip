@@ -597,6 +597,96 @@ __global__ __launch_bounds__(256, 2) void ec_bitmatrix_kernel(
   }
 }
 
+// Register-accumulator bitmatrix variant (v3): no LDS data staging, no
+// barrier — the matmul kernel's continuous-stream structure applied to
+// the bitmatrix map. Each thread owns one 16 B position v of a superword
+// sw and streams ALL k*w source packets once, XOR-ing each into the
+// NR = n_out*w output-row accumulators its column selects; column masks
+// are wave-uniform (readfirstlane -> scalar branch per statically
+// unrolled row, no divergence). Rationale: the LDS-window kernel's
+// barrier-phased bursts cap at ~4.9-5.1 TB/s while the matmul kernel
+// sustains 5.97 at the same 2.67:1 R/W mix — the mix is not the
+// ceiling, the burst structure is (membench ladder + bench r2).
+// NR <= 32 (n_out <= 4); larger n_out falls back to the LDS kernel.
+struct EcBitRegParams {
+  int n_src;
+  int pkt;
+  int src_ids[ECX_MAX_K];
+  int out_ids[4];
+  uint32_t colmask[ECX_MAX_K * 8];  // bit r of colmask[j*8+c]: row r uses (j,c)
+};
+
+template <bool NT, bool ACCUM, int NR>
+__global__ __launch_bounds__(256, 2) void ec_bitmatrix_reg_kernel(
+    const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
+    const uint8_t* __restrict__ blob, long chunk_bytes, int cps,
+    long n_pos) {
+  const EcBitRegParams* bp = (const EcBitRegParams*)blob;
+  __shared__ uint32_t s_cm[ECX_MAX_K * 8];
+  __shared__ int s_src[ECX_MAX_K];
+  __shared__ int s_out[4];
+  const int n_src = bp->n_src;
+  const int pkt = bp->pkt;
+  for (int t = threadIdx.x; t < n_src * 8; t += blockDim.x)
+    s_cm[t] = bp->colmask[t];
+  for (int t = threadIdx.x; t < n_src; t += blockDim.x)
+    s_src[t] = bp->src_ids[t];
+  for (int t = threadIdx.x; t < NR / 8; t += blockDim.x)
+    s_out[t] = bp->out_ids[t];
+  __syncthreads();
+
+  const long vp_sw = (long)pkt >> 4;  // 16B vecs per packet
+  const uint8_t* sbase = buf + (long)blockIdx.y * cps * chunk_bytes;
+  uint8_t* obase = obuf + (long)blockIdx.y * cps * chunk_bytes;
+
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < n_pos;
+       p += (long)gridDim.x * blockDim.x) {
+    const long sw = p / vp_sw;
+    const long v = p - sw * vp_sw;
+    const long sw_off = sw * 8 * (long)pkt + (v << 4);
+    v4u acc[NR];
+    if (ACCUM) {
+#pragma unroll
+      for (int r = 0; r < NR; r++)
+        acc[r] = *reinterpret_cast<const v4u*>(
+            obase + (long)s_out[r >> 3] * chunk_bytes + sw_off +
+            (long)(r & 7) * pkt);
+    } else {
+#pragma unroll
+      for (int r = 0; r < NR; r++) acc[r] = v4u{0, 0, 0, 0};
+    }
+    for (int j = 0; j < n_src; j++) {
+      const uint8_t* sj = sbase + (long)s_src[j] * chunk_bytes + sw_off;
+      const uint32_t* cmj = s_cm + j * 8;
+#pragma unroll
+      for (int c = 0; c < 8; c++) {
+        const v4u d =
+            NT ? __builtin_nontemporal_load(
+                     reinterpret_cast<const v4u*>(sj + (long)c * pkt))
+               : *reinterpret_cast<const v4u*>(sj + (long)c * pkt);
+        const uint32_t m8 = __builtin_amdgcn_readfirstlane(cmj[c]);
+#pragma unroll
+        for (int r = 0; r < NR; r++) {
+          if (m8 & (1u << r)) {
+            acc[r].x ^= d.x; acc[r].y ^= d.y;
+            acc[r].z ^= d.z; acc[r].w ^= d.w;
+          }
+        }
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < NR; r++) {
+      v4u* dst = reinterpret_cast<v4u*>(obase +
+                                        (long)s_out[r >> 3] * chunk_bytes +
+                                        sw_off + (long)(r & 7) * pkt);
+      if (NT)
+        __builtin_nontemporal_store(acc[r], dst);
+      else
+        *dst = acc[r];
+    }
+  }
+}
+
 // Software-pipelined bitmatrix variant: two LDS window buffers; each wave
 // issues its glds batch for window t+1, then waits ONLY for window t's
 // batch (counted s_waitcnt vmcnt(G) — vmcnt retires in order) and
@@ -1437,6 +1527,62 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   int vq_shift = 0;
   while ((1 << vq_shift) < q / 16) vq_shift++;
   if ((16 << vq_shift) != q) return ECX_ERR_INVAL;
+
+  // v3 register-accumulator path (no LDS staging, no barrier): n_out <= 4
+  // keeps the row accumulators within the register budget (NR*4 + temps
+  // VGPRs -> 3 waves/SIMD at NR=32); larger n_out uses the LDS kernel.
+  static const int env_reg = [] {
+    const char* v = getenv("ECX_BITREG");
+    return v ? atoi(v) : 0;
+  }();
+  if (env_reg && w == 8 && n_out <= 4 && (pkt & 15) == 0) {
+    EcBitRegParams hdr;
+    std::memset(&hdr, 0, sizeof(hdr));
+    hdr.n_src = n_src;
+    hdr.pkt = pkt;
+    for (int i = 0; i < n_src; i++) hdr.src_ids[i] = src_ids[i];
+    for (int j = 0; j < n_out; j++) hdr.out_ids[j] = out_ids[j];
+    const int W8 = n_src * 8, nr = n_out * 8;
+    for (int j = 0; j < n_src; j++)
+      for (int c = 0; c < 8; c++) {
+        uint32_t mbits = 0;
+        for (int r = 0; r < nr; r++)
+          if (bit_rows[(size_t)r * W8 + j * 8 + c]) mbits |= 1u << r;
+        hdr.colmask[j * 8 + c] = mbits;
+      }
+    int r = ensure_jobs(ctx, s, sizeof(hdr));
+    if (r != ECX_OK) return r;
+    HIP_TRY(wait_event(s.ev_jobs));
+    std::memcpy(s.h_jobs, &hdr, sizeof(hdr));
+    HIP_TRY(hipMemcpyAsync(s.d_jobs, s.h_jobs, sizeof(hdr),
+                           hipMemcpyHostToDevice, s.stream));
+    HIP_TRY(hipEventRecord(s.ev_jobs, s.stream));
+    const long n_pos = (long)(chunk_bytes >> 7);  // 16B vecs, 8 pkts/sw
+    dim3 grid((unsigned)std::min<long>((n_pos + 255) / 256, 16384),
+              (unsigned)n_stripes);
+    if (time_it) HIP_TRY(hipEventRecord(s.ev_start, s.stream));
+#define ECX_BRK(NR_)                                                    \
+  hipLaunchKernelGGL(                                                   \
+      (env_nt ? (accum ? ec_bitmatrix_reg_kernel<true, true, NR_>       \
+                       : ec_bitmatrix_reg_kernel<true, false, NR_>)     \
+              : (accum ? ec_bitmatrix_reg_kernel<false, true, NR_>      \
+                       : ec_bitmatrix_reg_kernel<false, false, NR_>)),  \
+      grid, dim3(256), 0, s.stream, d_buf, d_obuf, s.d_jobs,            \
+      (long)chunk_bytes, ctx->k + ctx->m, n_pos)
+    switch (n_out) {
+      case 1: ECX_BRK(8); break;
+      case 2: ECX_BRK(16); break;
+      case 3: ECX_BRK(24); break;
+      default: ECX_BRK(32); break;
+    }
+#undef ECX_BRK
+    HIP_TRY(hipGetLastError());
+    if (time_it) {
+      HIP_TRY(hipEventRecord(s.ev_stop, s.stream));
+      s.timed = true;
+    }
+    return ECX_OK;
+  }
 
   // build blob: header + ops
   const int n_rows = n_out * w, W = n_src * w;
