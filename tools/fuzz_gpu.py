@@ -28,15 +28,18 @@ def main():
     # reed_sol_r6_op is a harness-level alias of the isa RS-van matrix at
     # m=2 (tested in test_harness*); the C-ABI exposes the 5 base ids
     techs = ["reed_sol_van", "cauchy", "jerasure_reed_sol_van",
-             "jerasure_reed_sol_van_w16", "cauchy_orig"]
+             "jerasure_reed_sol_van_w16", "cauchy_orig", "cauchy_good"]
+    BITM = ("cauchy_orig", "cauchy_good")
     while time.monotonic() < stop:
         tech = techs[rng.integers(0, len(techs))]
-        if tech == "cauchy_orig":
+        if tech in BITM:
             k, m = int(rng.integers(2, 13)), int(rng.integers(1, 5))
+            if tech == "cauchy_good" and m == 2:
+                m = 3  # m=2 refused (cbest tables unsourceable)
         else:
             k, m = int(rng.integers(2, 21)), int(rng.integers(1, 5))
         n = k + m
-        if tech == "cauchy_orig":
+        if tech in BITM:
             pkt = int(rng.choice([512, 2048]))
             sw = 8 * pkt
             C = sw * int(rng.integers(1, 9))
@@ -53,8 +56,9 @@ def main():
                     rng.integers(0, 256, C, dtype=np.uint8)
                     for _ in range(k)]
             full = [np.zeros(C, np.uint8) if d is None else d for d in data]
-            if tech == "cauchy_orig":
-                want = oracle.bitmatrix_encode(k, m, full, pkt)
+            if tech in BITM:
+                want = oracle.bitmatrix_encode(k, m, full, pkt,
+                                               technique=tech)
             elif tech == "jerasure_reed_sol_van_w16":
                 want = oracle.encode_w16(k, m, full)
             else:

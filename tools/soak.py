@@ -41,8 +41,9 @@ def main():
         try:
             data = [rng.integers(0, 256, C, dtype=np.uint8)
                     for _ in range(k)]
-            if tech == "cauchy_orig":
-                want = oracle.bitmatrix_encode(k, m, data, 2048)
+            if tech in ("cauchy_orig", "cauchy_good"):
+                want = oracle.bitmatrix_encode(k, m, data, 2048,
+                                               technique=tech)
             elif tech == "jerasure_reed_sol_van_w16":
                 want = oracle.encode_w16(k, m, data)
             else:
@@ -106,7 +107,7 @@ def main():
             ("reed_sol_van", 8, 3), ("cauchy", 6, 2),
             ("jerasure_reed_sol_van", 4, 2),
             ("jerasure_reed_sol_van_w16", 5, 3),
-            ("cauchy_orig", 4, 2)])
+            ("cauchy_orig", 4, 2), ("cauchy_good", 4, 3)])
     ] + [threading.Thread(target=batch_worker)]
     for w in workers:
         w.start()
