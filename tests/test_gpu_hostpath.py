@@ -127,3 +127,59 @@ def test_all_zeros_sentinels_with_explicit_chunk_bytes(mods):
             assert p.nbytes == 65536 and not p.any()
     finally:
         ctx.close()
+
+
+def test_graph_replay_across_size_changes(mods):
+    """Regression for the round-2 GPU memory fault: single-tile host
+    calls replay captured hipGraphs that bake in the pinned/device
+    staging addresses; growing the pipe (a bigger call on the same slot)
+    reallocates those buffers and must invalidate the cached graphs.
+    Alternate small and large calls on ONE context (n_streams=1 pins
+    every call to the same slot) and byte-check each against the
+    oracle."""
+    ceph_amd, oracle = mods
+    k, m = 4, 2
+    ctx = ceph_amd.EcContext(k, m, "reed_sol_van", device=0, n_streams=1)
+    rng = np.random.default_rng(0x6F)
+    try:
+        for C in (64 << 10, 64 << 10, 1 << 20, 64 << 10, 2 << 20,
+                  128 << 10, 64 << 10):
+            data = [rng.integers(0, 256, C, dtype=np.uint8)
+                    for _ in range(k)]
+            par = ctx.encode_chunks(data)
+            want = oracle.encode("reed_sol_van", k, m, data)
+            for j in range(m):
+                assert np.array_equal(par[j], want[j]), (C, j)
+    finally:
+        ctx.close()
+
+
+def test_small_calls_many_shapes_one_slot(mods):
+    """Graph cache keying: different (n_src, n_out, tl) combinations on
+    one slot (decode plans vary n_src/n_out) must not cross-talk."""
+    ceph_amd, oracle = mods
+    rng = np.random.default_rng(0x51A)
+    for (k, m) in ((4, 2), (6, 3), (4, 2)):
+        ctx = ceph_amd.EcContext(k, m, "reed_sol_van", device=0,
+                                 n_streams=1)
+        try:
+            C = 64 << 10
+            data = [rng.integers(0, 256, C, dtype=np.uint8)
+                    for _ in range(k)]
+            par = ctx.encode_chunks(data)
+            want = oracle.encode("reed_sol_van", k, m, data)
+            for j in range(m):
+                assert np.array_equal(par[j], want[j])
+            # decode with 1..m erasures => varying (n_src, n_out) keys
+            for ne in range(1, m + 1):
+                chunks = [d.copy() for d in data] + [p.copy() for p in par]
+                er = list(range(ne))
+                present = [i not in er for i in range(k + m)]
+                for e in er:
+                    chunks[e][:] = 0
+                ctx.decode_chunks(chunks, present)
+                ref = data + par
+                for i in range(k + m):
+                    assert np.array_equal(chunks[i], ref[i]), (k, m, ne, i)
+        finally:
+            ctx.close()
