@@ -217,6 +217,25 @@ class ErasureCodeOracle final : public ErasureCode {
       if ((int)datashard >= k_) continue;
       for (auto &&[codingshard, codingbuf] : out) {
         if ((int)codingshard < k_) continue;
+        if (is_bitmatrix()) {
+          // schedule-delta semantics (schedule_apply_delta filtered to
+          // one (s, d) pair, ErasureCodeJerasure.cc:348-377): the pair's
+          // w x w bitmatrix block applied per superword, XOR into parity
+          const int i = (int)codingshard - k_, j = (int)datashard;
+          const int W = k_ * w_;
+          const size_t sw = (size_t)w_ * packetsize_;
+          uint8_t *cp = const_cast<uint8_t *>(codingbuf.c_str());
+          const uint8_t *dp = databuf.c_str();
+          for (size_t off = 0; off + sw <= codingbuf.length(); off += sw)
+            for (int r = 0; r < w_; r++)
+              for (int c = 0; c < w_; c++)
+                if (bitmat_[(size_t)(i * w_ + r) * W + j * w_ + c])
+                  ecref_xor_region(cp + off + (size_t)r * packetsize_,
+                                   dp + off + (size_t)c * packetsize_,
+                                   cp + off + (size_t)r * packetsize_,
+                                   packetsize_);
+          continue;
+        }
         uint8_t c = gen_[(size_t)(int)codingshard * k_ + (int)datashard];
         ecref_region_mul_xor(c, databuf.c_str(),
                              const_cast<uint8_t *>(codingbuf.c_str()),

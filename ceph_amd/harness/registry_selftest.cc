@@ -139,6 +139,48 @@ int main(int argc, char **argv) {
     CHECK(same, "parity delta == re-encode");
   }
   {
+    // bitmatrix (cauchy_orig/cauchy_good) parity delta == re-encode:
+    // schedule-delta semantics on CPU via the oracle fixture
+    // (schedule_apply_delta, ErasureCodeJerasure.cc:348-377)
+    for (const char *tech : {"cauchy_orig", "cauchy_good"}) {
+      ErasureCodeProfile p{{"k", "4"}, {"m", "3"}, {"technique", tech},
+                           {"packetsize", "128"}};
+      ErasureCodeInterfaceRef ec3;
+      int r = reg.factory("oracle", dir, p, &ec3, &ss);
+      CHECK(r == 0, std::string("oracle ") + tech + " factory");
+      if (r) continue;
+      const unsigned C = 8 * 128 * 2;  // 2 superwords
+      std::mt19937_64 rng(7);
+      shard_id_map<buffer> in(7), out(7);
+      for (int i = 0; i < 4; i++) {
+        buffer b = buffer::create_aligned(C);
+        for (size_t w = 0; w < C / 8; w++)
+          ((uint64_t *)b.c_str())[w] = rng();
+        in[i] = b;
+      }
+      for (int j = 4; j < 7; j++) out[j] = buffer::create_aligned(C);
+      CHECK(ec3->encode_chunks(in, out) == 0,
+            std::string(tech) + " encode");
+      buffer newc = buffer::create_aligned(C);
+      for (size_t w = 0; w < C / 8; w++)
+        ((uint64_t *)newc.c_str())[w] = rng();
+      buffer delta = buffer::create_aligned(C);
+      ec3->encode_delta(in.at(2), newc, &delta);
+      shard_id_map<buffer> din(7), dout(7);
+      din[2] = delta;
+      for (int j = 4; j < 7; j++) dout[j] = out.at(j);
+      ec3->apply_delta(din, dout);
+      in[2] = newc;
+      shard_id_map<buffer> out2(7);
+      for (int j = 4; j < 7; j++) out2[j] = buffer::create_aligned(C);
+      ec3->encode_chunks(in, out2);
+      bool same = true;
+      for (int j = 4; j < 7; j++)
+        same &= !std::memcmp(out.at(j).c_str(), out2.at(j).c_str(), C);
+      CHECK(same, std::string(tech) + " schedule delta == re-encode");
+    }
+  }
+  {
     // minimum_to_decode semantics (ErasureCode.cc:154-170)
     shard_id_set want, avail, minimum;
     want.insert(0);
