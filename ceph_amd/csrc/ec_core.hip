@@ -2596,7 +2596,15 @@ static int pipelined_matmul_host(ecx_ctx* ctx, Slot& s,
     const char* e = getenv("ECX_HGRAPH");
     return e ? atoi(e) : 1;
   }();
-  if (T == 1 && env_hg) {
+  // Graph replay wins where per-call submit overhead dominates and loses
+  // where hipGraphLaunch's runtime serialisation bites (measured, one
+  // box: 64 KiB x1 +25%, 64 KiB x8 callers -12%, 256 KiB x1 -9%,
+  // 1 MiB x8 -21%). Default: small calls only.
+  static const size_t env_hg_max = [] {
+    const char* e = getenv("ECX_HGRAPH_MAX");
+    return e ? (size_t)atol(e) : (size_t)(128 << 10);
+  }();
+  if (T == 1 && env_hg && chunk_bytes <= env_hg_max) {
     const size_t tl = chunk_bytes;
     uint8_t* hbuf = s.h_pipe;
     uint8_t* dbuf = s.d_pipe;
