@@ -526,3 +526,46 @@ def test_matmul_batch_arbitrary_rows():
     finally:
         ctx.dbuf_free(d)
         ctx.close()
+
+
+@pytest.mark.parametrize("tech,k,m", [
+    ("reed_sol_van", 12, 4), ("cauchy", 12, 4),
+    ("jerasure_reed_sol_van", 12, 4), ("cauchy_orig", 12, 4),
+    ("cauchy_good", 12, 4), ("cauchy_good", 10, 4),
+])
+def test_sampled_decode_headline_shapes(tech, k, m):
+    """Decode parity at the BASELINE headline widths (k=12/10, m=4):
+    exhaustive single-erasure plus 60 sampled multi-erasure patterns
+    (the (16 choose e) spaces are too large to sweep on every run)."""
+    from itertools import combinations
+    bitm = tech in ("cauchy_orig", "cauchy_good")
+    p = 512
+    C = (8 * p * 2) if bitm else 64 * 1024
+    n = k + m
+    rng = np.random.default_rng(0xBEEF ^ (k << 8) ^ m)
+    kw = {"packetsize": p} if bitm else {}
+    ctx = ceph_amd.EcContext(k, m, tech, device=0, **kw)
+    try:
+        data = [rng.integers(0, 256, C, dtype=np.uint8) for _ in range(k)]
+        par = ctx.encode_chunks(data)
+        if bitm:
+            want = oracle.bitmatrix_encode(k, m, data, p, technique=tech)
+        else:
+            want = oracle.encode(tech, k, m, data)
+        for j in range(m):
+            assert np.array_equal(par[j], want[j]), ("enc", j)
+        full = data + par
+        patterns = [(e,) for e in range(n)]
+        for _ in range(60):
+            e = int(rng.integers(2, m + 1))
+            patterns.append(tuple(sorted(
+                rng.choice(n, size=e, replace=False).tolist())))
+        for er in patterns:
+            present = [i not in er for i in range(n)]
+            chunks = [c.copy() if present[i] else np.zeros(C, np.uint8)
+                      for i, c in enumerate(full)]
+            ctx.decode_chunks(chunks, present)
+            for i in range(n):
+                assert np.array_equal(chunks[i], full[i]), (er, i)
+    finally:
+        ctx.close()
