@@ -1528,14 +1528,19 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   while ((1 << vq_shift) < q / 16) vq_shift++;
   if ((16 << vq_shift) != q) return ECX_ERR_INVAL;
 
-  // v3 register-accumulator path (no LDS staging, no barrier): n_out <= 4
-  // keeps the row accumulators within the register budget (NR*4 + temps
-  // VGPRs -> 3 waves/SIMD at NR=32); larger n_out uses the LDS kernel.
+  // v3 register-accumulator path (no LDS staging, no barrier): measured
+  // policy — for n_out <= 2 (single/double-erasure decode, delta apply)
+  // the LDS kernel idles most of each block's compute items while v3
+  // keeps every lane busy (+7% at n_out=1: 7.13 vs 7.64 ms); at
+  // n_out >= 3 the LDS kernel wins (10.43 vs 10.85 ms at m=3 encode).
+  // ECX_BITREG: 0 = never, 1 = always (n_out <= 4), 2/unset = auto.
   static const int env_reg = [] {
     const char* v = getenv("ECX_BITREG");
-    return v ? atoi(v) : 0;
+    return v ? atoi(v) : 2;
   }();
-  if (env_reg && w == 8 && n_out <= 4 && (pkt & 15) == 0) {
+  const bool use_reg =
+      env_reg == 1 ? n_out <= 4 : (env_reg == 2 ? n_out <= 2 : false);
+  if (use_reg && w == 8 && (pkt & 15) == 0) {
     EcBitRegParams hdr;
     std::memset(&hdr, 0, sizeof(hdr));
     hdr.n_src = n_src;
