@@ -1529,17 +1529,17 @@ static int run_bitmatrix(ecx_ctx* ctx, int slot_i, const uint8_t* d_buf,
   if ((16 << vq_shift) != q) return ECX_ERR_INVAL;
 
   // v3 register-accumulator path (no LDS staging, no barrier): measured
-  // policy — for n_out <= 2 (single/double-erasure decode, delta apply)
-  // the LDS kernel idles most of each block's compute items while v3
-  // keeps every lane busy (+7% at n_out=1: 7.13 vs 7.64 ms); at
-  // n_out >= 3 the LDS kernel wins (10.43 vs 10.85 ms at m=3 encode).
+  // policy — at n_out == 1 (single-erasure decode, delta apply) the LDS
+  // kernel idles most of each block's compute items while v3 keeps every
+  // lane busy (7.13 vs 7.64 ms); at n_out >= 2 the LDS kernel wins
+  // (n_out=2 decode 8.02 vs 11.04; m=3 encode 10.43 vs 10.85).
   // ECX_BITREG: 0 = never, 1 = always (n_out <= 4), 2/unset = auto.
   static const int env_reg = [] {
     const char* v = getenv("ECX_BITREG");
     return v ? atoi(v) : 2;
   }();
   const bool use_reg =
-      env_reg == 1 ? n_out <= 4 : (env_reg == 2 ? n_out <= 2 : false);
+      env_reg == 1 ? n_out <= 4 : (env_reg == 2 ? n_out == 1 : false);
   if (use_reg && w == 8 && (pkt & 15) == 0) {
     EcBitRegParams hdr;
     std::memset(&hdr, 0, sizeof(hdr));
