@@ -3,7 +3,12 @@
 
 Run once and commit the output. These vectors SELF-pin the oracle (catch
 regressions); byte-level parity vs compiled jerasure/isa-l remains an
-external spot-check (see oracle/ec_ref.h "PARITY PINNING STATUS")."""
+external spot-check (see oracle/ec_ref.h "PARITY PINNING STATUS").
+When a jerasure/ISA-L build or a real Ceph install becomes available,
+run `tools/corpus_check.sh <their-corpus-dir>` (one command, maps plugin
+and technique names) — and fold their chunks into tests/golden/corpus as
+externally-generated KATs, which upgrades the pinning status from
+"partial" for every technique it covers."""
 import os
 import sys
 

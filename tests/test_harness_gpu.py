@@ -150,3 +150,14 @@ def test_mi355x_flags_mirror_reference(profile, expected):
     r = run_bench("-p", "mi355x", *profile, "--flags")
     assert r.returncode == 0, r.stderr
     assert r.stdout.strip() == expected
+
+
+def test_cauchy_good_cli_exhaustive_double_erasure():
+    """cauchy_good through the dlopen plugin boundary at the headline
+    cauchy width (k=10 m=4): exhaustive 2-erasure decode with byte
+    verification by the CLI (benchmark.cc:211-258 style)."""
+    r = run_bench("-p", "mi355x", "-P", "technique=cauchy_good",
+                  "-P", "k=10", "-P", "m=4", "-P", "packetsize=512",
+                  "-s", str(10 * 8 * 512 * 2), "-i", "1", "-w", "decode",
+                  "-e", "2", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
