@@ -574,10 +574,14 @@ def main():
         import torch
         import torch.distributed as tdist
         # modulo so an N-rank run also works on fewer devices (single-box
-        # smoke tests); identity on the 8-GPU node
+        # smoke tests); identity on the 8-GPU node. ECX_BENCH_BACKEND=gloo
+        # lets a multi-rank run share ONE GPU (NCCL refuses duplicate
+        # devices) — used to validate the full distributed path end to
+        # end on a 1-GPU box; the driver's 8-GPU run keeps nccl/RCCL.
+        backend = os.environ.get("ECX_BENCH_BACKEND", "nccl")
         device = local_rank % max(1, torch.cuda.device_count())
         torch.cuda.set_device(device)
-        tdist.init_process_group("nccl")
+        tdist.init_process_group(backend)
         dist = tdist
 
     import ceph_amd
@@ -643,7 +647,8 @@ def main():
     if dist:
         import torch
         torch.cuda.synchronize()
-        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
         dist.barrier()
