@@ -2136,6 +2136,28 @@ int ecx_cauchy_n_ones_probe(int e) {
   return ecx::cauchy_n_ones((uint8_t)e);
 }
 
+int ecx_decode_rows_probe(int technique, int k, int m, uint64_t present_mask,
+                          int* survivors, int* erased, uint8_t* rows) {
+  // CPU-only probe of the decode-plan composition (survivor selection +
+  // submatrix inversion + per-erasure coefficient rows, gf.cpp
+  // compose_decode_rows — the math behind get_decode_plan's LRU).
+  // Fills survivors[k], erased[<=m], rows[n_erased*k]; returns n_erased
+  // or a negative errno. Lets CPU tests pin the plan math against the
+  // oracle's decode composition without a GPU context.
+  if (!survivors || !erased || !rows || k < 1 || m < 1 || k + m > 64)
+    return ECX_ERR_INVAL;
+  std::vector<uint8_t> gen;
+  if (!ecx::gen_matrix(technique, gen, k, m)) return ECX_ERR_INVAL;
+  std::vector<int> sv, er;
+  std::vector<uint8_t> rw;
+  if (!ecx::compose_decode_rows(gen, k, m, present_mask, sv, er, rw))
+    return ECX_ERR_IO;
+  for (int i = 0; i < k; i++) survivors[i] = sv[i];
+  for (size_t i = 0; i < er.size(); i++) erased[i] = er[i];
+  std::memcpy(rows, rw.data(), rw.size());
+  return (int)er.size();
+}
+
 int ecx_shec_matrix(int k, int m, int c, int single, uint8_t* out) {
   if (!out) return ECX_ERR_INVAL;
   std::vector<uint8_t> coding;

@@ -225,6 +225,11 @@ ECX_API int ecx_shec_matrix(int k, int m, int c, int single, uint8_t *out);
  * top — and jerasure cauchy.c's cauchy_n_ones(e, w=8). */
 ECX_API int ecx_gen_matrix_probe(int technique, int k, int m, uint8_t *out);
 ECX_API int ecx_cauchy_n_ones_probe(int e);
+/* decode-plan composition probe: survivors[k], erased[<=m],
+ * rows[n_erased*k]; returns n_erased or -errno. */
+ECX_API int ecx_decode_rows_probe(int technique, int k, int m,
+                                  uint64_t present_mask, int *survivors,
+                                  int *erased, uint8_t *rows);
 
 /* Generic device-batch GF(2^8) matmul over the standard batch layout:
  * out chunk ids = XOR_i rows[j*n_src+i] * src chunk ids, per stripe. The
