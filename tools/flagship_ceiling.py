@@ -16,7 +16,15 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np  # noqa: E402
 import ceph_amd  # noqa: E402
 
-K, M, C, S = 8, 3, 1 << 20, 4096
+import argparse
+
+ap = argparse.ArgumentParser()
+ap.add_argument("--k", type=int, default=8)
+ap.add_argument("--m", type=int, default=3)
+ap.add_argument("--technique", default="reed_sol_van")
+ap.add_argument("--stripes", type=int, default=4096)
+A = ap.parse_args()
+K, M, C, S = A.k, A.m, 1 << 20, A.stripes
 
 
 def run(ctx, dptr, steps=5):
@@ -28,7 +36,7 @@ def run(ctx, dptr, steps=5):
 
 
 def main():
-    ctx = ceph_amd.EcContext(K, M, "reed_sol_van", device=0)
+    ctx = ceph_amd.EcContext(K, M, A.technique, device=0)
     n = K + M
     buf = S * n * C
     d = ctx.dbuf_alloc(buf)
@@ -39,6 +47,7 @@ def main():
     ctx.set_matrix(np.ones((M, K), dtype=np.uint8))
     allones = run(ctx, d)
     print(json.dumps({
+        "k": K, "m": M, "technique": A.technique,
         "real_ms": round(real, 4), "real_GBs": round(alg / real / 1e6, 0),
         "allones_ms": round(allones, 4),
         "allones_GBs": round(alg / allones / 1e6, 0),
