@@ -91,7 +91,13 @@ __device__ __forceinline__ uint32_t ecx_gfmul4(uint32_t l0, uint32_t l1,
 // transaction.
 // KN > 0 statically unrolls the source loop (flagship k=8): all k loads
 // are grouped ahead of the compute within a position, deepening MLP.
-template <int NOUT, bool ACCUM, int VPT, bool NT, int KN = 0>
+// PF: one-ahead source prefetch — issue source i+1's load before
+// computing on source i, so each wave keeps a load in flight across the
+// compute body (the dynamic i-loop otherwise serialises load -> use).
+// Costs VPT*4 VGPRs; targeted at the NOUT=4 shapes where the all-ones
+// probe measured 15-20% of wall as VALU not hidden behind the stream.
+template <int NOUT, bool ACCUM, int VPT, bool NT, int KN = 0,
+          bool PF = false>
 __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const EcLaunchParams* __restrict__ pb, long chunk_bytes,
@@ -140,6 +146,20 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
         }
       }
 
+    uint32_t dnf[PF ? VPT : 1][4];
+    if (PF) {
+      const uint8_t* sp0 = sbase + (long)s_src[0] * chunk_bytes;
+#pragma unroll
+      for (int v = 0; v < VPT; v++) {
+        v4u d = {0, 0, 0, 0};
+        if (live[v]) {
+          const v4u* p4 = reinterpret_cast<const v4u*>(sp0 + off[v]);
+          d = NT ? __builtin_nontemporal_load(p4) : *p4;
+        }
+        dnf[v][0] = d.x; dnf[v][1] = d.y;
+        dnf[v][2] = d.z; dnf[v][3] = d.w;
+      }
+    }
 #pragma unroll (KN ? KN : 1)
     for (int i = 0; i < n_src; i++) {
       const uint8_t* sp = sbase + (long)s_src[i] * chunk_bytes;
@@ -148,11 +168,30 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
 #pragma unroll
       for (int v = 0; v < VPT; v++) {
         v4u d = {0, 0, 0, 0};
-        if (live[v]) {
+        if (PF) {
+          d.x = dnf[v][0]; d.y = dnf[v][1];
+          d.z = dnf[v][2]; d.w = dnf[v][3];
+        } else if (live[v]) {
           const v4u* p4 = reinterpret_cast<const v4u*>(sp + off[v]);
           d = NT ? __builtin_nontemporal_load(p4) : *p4;
         }
         dq[v][0] = d.x; dq[v][1] = d.y; dq[v][2] = d.z; dq[v][3] = d.w;
+      }
+      if (PF && i + 1 < n_src) {
+        const uint8_t* spn = sbase + (long)s_src[i + 1] * chunk_bytes;
+#pragma unroll
+        for (int v = 0; v < VPT; v++) {
+          v4u d = {0, 0, 0, 0};
+          if (live[v]) {
+            const v4u* p4 = reinterpret_cast<const v4u*>(spn + off[v]);
+            d = NT ? __builtin_nontemporal_load(p4) : *p4;
+          }
+          dnf[v][0] = d.x; dnf[v][1] = d.y;
+          dnf[v][2] = d.z; dnf[v][3] = d.w;
+        }
+      }
+#pragma unroll
+      for (int v = 0; v < VPT; v++) {
 #pragma unroll
         for (int q = 0; q < 4; q++) {
           i7l[v][q] = dq[v][q] & 0x07070707u;
@@ -1269,11 +1308,27 @@ static int matmul_dispatch(hipStream_t stream, const uint8_t* d_buf,
     const char* v = getenv("ECX_K8");
     return v ? atoi(v) : 0;
   }();
+  // one-ahead source prefetch (PF template): keeps a load in flight
+  // across each compute body. ECX_PF: 0 = never, 1 = always, 2/unset =
+  // auto (NOUT >= 4, where the all-ones probe measured 15-20% exposed
+  // VALU; the NOUT <= 3 shapes are already at their memory floor).
+  static const int env_pf = [] {
+    const char* v = getenv("ECX_PF");
+    return v ? atoi(v) : 2;
+  }();
+  const bool pf = env_pf == 1 || (env_pf == 2 && n_out >= 4);
   const bool k8 = env_k8 && !accum && vpt == 1 && n_src == 8;
 #define ECX_LAUNCH(NO, AC, VP, NTF)                                          \
-  hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF>), grid,           \
-                     dim3(256), 0, stream, d_buf, d_obuf, d_params,          \
-                     (long)chunk_bytes, cps, vecs)
+  do {                                                                       \
+    if (pf)                                                                  \
+      hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF, 0, true>),    \
+                         grid, dim3(256), 0, stream, d_buf, d_obuf,          \
+                         d_params, (long)chunk_bytes, cps, vecs);            \
+    else                                                                     \
+      hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF>), grid,       \
+                         dim3(256), 0, stream, d_buf, d_obuf, d_params,      \
+                         (long)chunk_bytes, cps, vecs);                      \
+  } while (0)
 #define ECX_LAUNCH_K8(NO, NTF)                                               \
   hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, false, 1, NTF, 8>), grid,      \
                      dim3(256), 0, stream, d_buf, d_obuf, d_params,          \
