@@ -91,13 +91,12 @@ __device__ __forceinline__ uint32_t ecx_gfmul4(uint32_t l0, uint32_t l1,
 // transaction.
 // KN > 0 statically unrolls the source loop (flagship k=8): all k loads
 // are grouped ahead of the compute within a position, deepening MLP.
-// PF: one-ahead source prefetch — issue source i+1's load before
-// computing on source i, so each wave keeps a load in flight across the
-// compute body (the dynamic i-loop otherwise serialises load -> use).
-// Costs VPT*4 VGPRs; targeted at the NOUT=4 shapes where the all-ones
+// PF: source prefetch depth — issue source i+PF's load before computing
+// on source i, so each wave keeps PF loads in flight across the compute
+// body (the dynamic i-loop otherwise serialises load -> use). Costs
+// PF*VPT*4 VGPRs; targeted at the NOUT=4 shapes where the all-ones
 // probe measured 15-20% of wall as VALU not hidden behind the stream.
-template <int NOUT, bool ACCUM, int VPT, bool NT, int KN = 0,
-          bool PF = false>
+template <int NOUT, bool ACCUM, int VPT, bool NT, int KN = 0, int PF = 0>
 __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
     const uint8_t* __restrict__ buf, uint8_t* __restrict__ obuf,
     const EcLaunchParams* __restrict__ pb, long chunk_bytes,
@@ -146,9 +145,11 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
         }
       }
 
-    uint32_t dnf[PF ? VPT : 1][4];
-    if (PF) {
-      const uint8_t* sp0 = sbase + (long)s_src[0] * chunk_bytes;
+    uint32_t dnf[PF ? PF : 1][VPT][4];
+#pragma unroll
+    for (int pd = 0; pd < PF; pd++) {
+      const int si = pd < n_src ? pd : n_src - 1;
+      const uint8_t* sp0 = sbase + (long)s_src[si] * chunk_bytes;
 #pragma unroll
       for (int v = 0; v < VPT; v++) {
         v4u d = {0, 0, 0, 0};
@@ -156,8 +157,8 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
           const v4u* p4 = reinterpret_cast<const v4u*>(sp0 + off[v]);
           d = NT ? __builtin_nontemporal_load(p4) : *p4;
         }
-        dnf[v][0] = d.x; dnf[v][1] = d.y;
-        dnf[v][2] = d.z; dnf[v][3] = d.w;
+        dnf[pd][v][0] = d.x; dnf[pd][v][1] = d.y;
+        dnf[pd][v][2] = d.z; dnf[pd][v][3] = d.w;
       }
     }
 #pragma unroll (KN ? KN : 1)
@@ -169,25 +170,36 @@ __global__ __launch_bounds__(256, 2) void ec_gf_matmul_kernel(
       for (int v = 0; v < VPT; v++) {
         v4u d = {0, 0, 0, 0};
         if (PF) {
-          d.x = dnf[v][0]; d.y = dnf[v][1];
-          d.z = dnf[v][2]; d.w = dnf[v][3];
+          d.x = dnf[0][v][0]; d.y = dnf[0][v][1];
+          d.z = dnf[0][v][2]; d.w = dnf[0][v][3];
         } else if (live[v]) {
           const v4u* p4 = reinterpret_cast<const v4u*>(sp + off[v]);
           d = NT ? __builtin_nontemporal_load(p4) : *p4;
         }
         dq[v][0] = d.x; dq[v][1] = d.y; dq[v][2] = d.z; dq[v][3] = d.w;
       }
-      if (PF && i + 1 < n_src) {
-        const uint8_t* spn = sbase + (long)s_src[i + 1] * chunk_bytes;
+      if (PF) {
+        // rotate the prefetch ring and issue source i+PF
 #pragma unroll
-        for (int v = 0; v < VPT; v++) {
-          v4u d = {0, 0, 0, 0};
-          if (live[v]) {
-            const v4u* p4 = reinterpret_cast<const v4u*>(spn + off[v]);
-            d = NT ? __builtin_nontemporal_load(p4) : *p4;
+        for (int pd = 0; pd + 1 < PF; pd++)
+#pragma unroll
+          for (int v = 0; v < VPT; v++)
+#pragma unroll
+            for (int q = 0; q < 4; q++) dnf[pd][v][q] = dnf[pd + 1][v][q];
+        if (i + PF < n_src) {
+          const uint8_t* spn = sbase + (long)s_src[i + PF] * chunk_bytes;
+#pragma unroll
+          for (int v = 0; v < VPT; v++) {
+            v4u d = {0, 0, 0, 0};
+            if (live[v]) {
+              const v4u* p4 = reinterpret_cast<const v4u*>(spn + off[v]);
+              d = NT ? __builtin_nontemporal_load(p4) : *p4;
+            }
+            dnf[PF ? PF - 1 : 0][v][0] = d.x;
+            dnf[PF ? PF - 1 : 0][v][1] = d.y;
+            dnf[PF ? PF - 1 : 0][v][2] = d.z;
+            dnf[PF ? PF - 1 : 0][v][3] = d.w;
           }
-          dnf[v][0] = d.x; dnf[v][1] = d.y;
-          dnf[v][2] = d.z; dnf[v][3] = d.w;
         }
       }
 #pragma unroll
@@ -1316,12 +1328,20 @@ static int matmul_dispatch(hipStream_t stream, const uint8_t* d_buf,
     const char* v = getenv("ECX_PF");
     return v ? atoi(v) : 2;
   }();
-  const bool pf = env_pf == 1 || (env_pf == 2 && n_out >= 4);
+  // depth: ECX_PF 0 = off, 1 = always depth-1, 2/unset = auto (depth 1
+  // at NOUT >= 4), 3 = always depth-2 (experiment)
+  const int pf = env_pf == 1 ? 1
+                 : env_pf == 3 ? 2
+                 : (env_pf == 2 && n_out >= 4) ? 1 : 0;
   const bool k8 = env_k8 && !accum && vpt == 1 && n_src == 8;
 #define ECX_LAUNCH(NO, AC, VP, NTF)                                          \
   do {                                                                       \
-    if (pf)                                                                  \
-      hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF, 0, true>),    \
+    if (pf == 2)                                                             \
+      hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF, 0, 2>),       \
+                         grid, dim3(256), 0, stream, d_buf, d_obuf,          \
+                         d_params, (long)chunk_bytes, cps, vecs);            \
+    else if (pf == 1)                                                        \
+      hipLaunchKernelGGL((ec_gf_matmul_kernel<NO, AC, VP, NTF, 0, 1>),       \
                          grid, dim3(256), 0, stream, d_buf, d_obuf,          \
                          d_params, (long)chunk_bytes, cps, vecs);            \
     else                                                                     \
