@@ -161,3 +161,31 @@ def test_cauchy_good_cli_exhaustive_double_erasure():
                   "-s", str(10 * 8 * 512 * 2), "-i", "1", "-w", "decode",
                   "-e", "2", "-E", "exhaustive")
     assert r.returncode == 0, r.stderr + r.stdout
+
+
+def test_lrc_headline_shape_gpu():
+    """LRC at the BASELINE configs[3]-adjacent shape (k=9 m=3 l=4 — the
+    reference's parse_kml rejects k=8; see DESIGN) through the dlopen
+    boundary: exhaustive single-erasure decode, byte-verified."""
+    r = run_bench("-p", "lrc", "-P", "k=9", "-P", "m=3", "-P", "l=4",
+                  "-s", str(9 * 65536), "-i", "1", "-w", "decode",
+                  "-e", "1", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
+
+
+def test_clay_wider_shape_gpu():
+    """Clay at a wider shape (k=6 m=3 d=8): two-erasure exhaustive decode
+    through decode_layered with the mi355x sub-codec."""
+    r = run_bench("-p", "clay", "-P", "k=6", "-P", "m=3", "-P", "d=8",
+                  "-s", str(6 * 65536), "-i", "1", "-w", "decode",
+                  "-e", "2", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
+
+
+def test_shec_wider_shape_gpu():
+    """SHEC k=8 m=4 c=3: exhaustive single-erasure decode (the widest
+    in-tree-documented shingle density at k=8)."""
+    r = run_bench("-p", "shec", "-P", "k=8", "-P", "m=4", "-P", "c=3",
+                  "-s", str(8 * 65536), "-i", "1", "-w", "decode",
+                  "-e", "1", "-E", "exhaustive")
+    assert r.returncode == 0, r.stderr + r.stdout
