@@ -31,6 +31,5 @@ ctx.close()
 PY
 done
 
-timeout 500 python tools/fuzz_gpu.py --seconds 420 --seed 0xR2 2>/dev/null \
-  || timeout 500 python tools/fuzz_gpu.py --seconds 420 --seed 777 2>&1 | tail -1
+timeout 500 python tools/fuzz_gpu.py --seconds 420 --seed 777 2>&1 | tail -1
 timeout 500 python tools/soak.py --seconds 400 2>&1 | tail -3
